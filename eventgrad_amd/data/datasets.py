@@ -1,0 +1,151 @@
+"""Dataset implementations (map-style: __len__ / __getitem__ -> (x, y))."""
+
+from __future__ import annotations
+
+import gzip
+import os
+import pickle
+import struct
+from typing import Optional, Tuple
+
+import numpy as np
+import torch
+
+from ..config import DataConfig
+
+MNIST_MEAN, MNIST_STD = 0.1307, 0.3081  # cent.cpp:55 Normalize
+
+
+class SyntheticImages:
+    """Deterministic learnable synthetic image classification.
+
+    Each class c has a fixed random prototype P_c ~ N(0,1); sample i of class
+    c = i % num_classes is P_c + noise * N(0,1) with per-index deterministic
+    noise. Classes are linearly separable for small noise, so optimization /
+    trigger dynamics resemble real training while needing no downloads.
+    """
+
+    def __init__(self, shape: Tuple[int, int, int], n: int,
+                 num_classes: int = 10, noise: float = 0.5, seed: int = 1234):
+        self.shape, self.n, self.num_classes = shape, n, num_classes
+        self.noise = noise
+        g = torch.Generator().manual_seed(seed)
+        self.prototypes = torch.randn((num_classes,) + shape, generator=g)
+        self.seed = seed
+
+    def __len__(self):
+        return self.n
+
+    def __getitem__(self, i: int):
+        i = int(i)
+        c = i % self.num_classes
+        g = torch.Generator().manual_seed(self.seed * 1000003 + i)
+        x = self.prototypes[c] + self.noise * torch.randn(self.shape,
+                                                          generator=g)
+        return x, c
+
+    def batch(self, idx: list) -> Tuple[torch.Tensor, torch.Tensor]:
+        xs, ys = zip(*(self[i] for i in idx))
+        return torch.stack(xs), torch.tensor(ys, dtype=torch.long)
+
+
+def _read_idx(path: str) -> np.ndarray:
+    op = gzip.open if path.endswith(".gz") else open
+    with op(path, "rb") as f:
+        data = f.read()
+    zeros, dtype, ndim = data[0], data[2], data[3]
+    assert zeros == 0 and dtype == 8, "unsupported IDX file"
+    dims = struct.unpack(f">{ndim}I", data[4:4 + 4 * ndim])
+    return np.frombuffer(data, dtype=np.uint8,
+                         offset=4 + 4 * ndim).reshape(dims)
+
+
+class MnistDataset:
+    """MNIST from the standard IDX files, normalized as the reference
+    (Normalize(0.1307, 0.3081), cent.cpp:55)."""
+
+    FILES = {
+        True: ("train-images-idx3-ubyte", "train-labels-idx1-ubyte"),
+        False: ("t10k-images-idx3-ubyte", "t10k-labels-idx1-ubyte"),
+    }
+
+    def __init__(self, root: str, train: bool = True):
+        imgf, lblf = self.FILES[train]
+        for suffix in ("", ".gz"):
+            p = os.path.join(root, imgf + suffix)
+            if os.path.exists(p):
+                imgf, lblf = imgf + suffix, lblf + suffix
+                break
+        self.images = _read_idx(os.path.join(root, imgf))
+        self.labels = _read_idx(os.path.join(root, lblf))
+
+    def __len__(self):
+        return len(self.labels)
+
+    def __getitem__(self, i: int):
+        x = torch.from_numpy(self.images[i].astype(np.float32) / 255.0)
+        x = (x - MNIST_MEAN) / MNIST_STD
+        return x.unsqueeze(0), int(self.labels[i])
+
+    def batch(self, idx: list):
+        x = torch.from_numpy(
+            self.images[idx].astype(np.float32) / 255.0).unsqueeze(1)
+        x = (x - MNIST_MEAN) / MNIST_STD
+        y = torch.from_numpy(self.labels[idx].astype(np.int64))
+        return x, y
+
+
+class Cifar10Dataset:
+    """CIFAR-10 from the standard binary batches (data_batch_*.bin) or the
+    python pickle distribution (data_batch_* pickles)."""
+
+    def __init__(self, root: str, train: bool = True):
+        imgs, lbls = [], []
+        names = ([f"data_batch_{i}" for i in range(1, 6)] if train
+                 else ["test_batch"])
+        for n in names:
+            binp = os.path.join(root, n + ".bin")
+            pkl = os.path.join(root, n)
+            if os.path.exists(binp):
+                raw = np.fromfile(binp, dtype=np.uint8).reshape(-1, 3073)
+                lbls.append(raw[:, 0].astype(np.int64))
+                imgs.append(raw[:, 1:].reshape(-1, 3, 32, 32))
+            elif os.path.exists(pkl):
+                with open(pkl, "rb") as f:
+                    d = pickle.load(f, encoding="bytes")
+                lbls.append(np.asarray(d[b"labels"], dtype=np.int64))
+                imgs.append(np.asarray(d[b"data"],
+                                       dtype=np.uint8).reshape(-1, 3, 32, 32))
+            else:
+                raise FileNotFoundError(f"CIFAR-10 batch not found: {binp}")
+        self.images = np.concatenate(imgs)
+        self.labels = np.concatenate(lbls)
+
+    def __len__(self):
+        return len(self.labels)
+
+    def __getitem__(self, i: int):
+        x = torch.from_numpy(self.images[i].astype(np.float32) / 255.0)
+        return x, int(self.labels[i])
+
+    def batch(self, idx: list):
+        x = torch.from_numpy(self.images[idx].astype(np.float32) / 255.0)
+        y = torch.from_numpy(self.labels[idx])
+        return x, y
+
+
+_SHAPES = {"mnist": (1, 28, 28), "cifar10": (3, 32, 32)}
+
+
+def build_dataset(cfg: DataConfig, train: bool):
+    name = cfg.dataset
+    if name in ("mnist", "cifar10") and cfg.data_path:
+        cls = MnistDataset if name == "mnist" else Cifar10Dataset
+        return cls(cfg.data_path, train)
+    # synthetic fallback (no-network environment)
+    shape = _SHAPES.get(name, _SHAPES["cifar10"])
+    if name == "synthetic-mnist":
+        shape = _SHAPES["mnist"]
+    n = cfg.synthetic_train_samples if train else cfg.synthetic_test_samples
+    return SyntheticImages(shape, n, cfg.num_classes, cfg.synthetic_noise,
+                           seed=1234 if train else 4321)

@@ -1,0 +1,140 @@
+"""CIFAR ResNet family re-implemented from dcifar10/common/resnet.hpp.
+
+Structure (resnet.hpp:111-157): 3x3 stem conv 3->64 (stride 1, no maxpool —
+the CIFAR variant), 4 stages at 64/128/256/512 channels with strides
+1/2/2/2, avg_pool2d(4), fc. BasicBlock = 2x(3x3 conv + BN), expansion 1
+(resnet.hpp:11-54); BottleNeck = 1-3-1 convs, expansion 4 (resnet.hpp:56-109);
+1x1-conv+BN downsampler when shape changes; all convs bias-free
+(resnet.hpp:3-9).
+
+The reference's ``make_layer`` pushes one block with the stride/downsampler
+and then ``blocks`` MORE (resnet.hpp:172-178), so each stage has layers[i]+1
+blocks: "ResNet18" = ResNet<BasicBlock>({2,2,2,2}) actually has 12 basic
+blocks, 17,444,682 parameters in 86 named tensors. We keep
+that behaviour under the ``quirk=True`` models (``resnet18q`` — the
+configuration the reference's CIFAR-10 baselines and message counts are
+quoted on) and also provide standard counts (``resnet18``).
+"""
+
+from __future__ import annotations
+
+import torch
+from torch import nn
+
+from ..ops import functional as O
+from .layers import BatchNorm2d, Conv2d
+
+
+def conv_op(in_ch, out_ch, k, stride, padding):
+    # resnet.hpp:3-9 — bias-free conv
+    return Conv2d(in_ch, out_ch, k, stride=stride, padding=padding, bias=False)
+
+
+class Downsampler(nn.Module):
+    """1x1 conv + BN shortcut (resnet.hpp:166-171)."""
+
+    def __init__(self, in_ch, out_ch, stride):
+        super().__init__()
+        self.conv = conv_op(in_ch, out_ch, 1, stride, 0)
+        self.bn = BatchNorm2d(out_ch)
+
+    def forward(self, x):
+        return self.bn(self.conv(x))
+
+
+class BasicBlock(nn.Module):
+    expansion = 1
+
+    def __init__(self, in_ch, out_ch, stride=1, downsample: bool = False):
+        super().__init__()
+        self.conv1 = conv_op(in_ch, out_ch, 3, stride, 1)
+        self.bn1 = BatchNorm2d(out_ch)
+        self.conv2 = conv_op(out_ch, out_ch, 3, 1, 1)
+        self.bn2 = BatchNorm2d(out_ch)
+        self.downsampler = (Downsampler(in_ch, out_ch * self.expansion, stride)
+                            if downsample else None)
+
+    def forward(self, x):
+        out = self.bn1(self.conv1(x), fuse_relu=True)
+        out = self.bn2(self.conv2(out))
+        residual = self.downsampler(x) if self.downsampler is not None else x
+        return O.add_relu(out, residual)
+
+
+class BottleNeck(nn.Module):
+    expansion = 4
+
+    def __init__(self, in_ch, out_ch, stride=1, downsample: bool = False):
+        super().__init__()
+        self.conv1 = conv_op(in_ch, out_ch, 1, 1, 0)
+        self.bn1 = BatchNorm2d(out_ch)
+        self.conv2 = conv_op(out_ch, out_ch, 3, stride, 1)
+        self.bn2 = BatchNorm2d(out_ch)
+        self.conv3 = conv_op(out_ch, out_ch * self.expansion, 1, 1, 0)
+        self.bn3 = BatchNorm2d(out_ch * self.expansion)
+        self.downsampler = (Downsampler(in_ch, out_ch * self.expansion, stride)
+                            if downsample else None)
+
+    def forward(self, x):
+        out = self.bn1(self.conv1(x), fuse_relu=True)
+        out = self.bn2(self.conv2(out), fuse_relu=True)
+        out = self.bn3(self.conv3(out))
+        residual = self.downsampler(x) if self.downsampler is not None else x
+        return O.add_relu(out, residual)
+
+
+class ResNet(nn.Module):
+    def __init__(self, block, layers, num_classes=10, quirk=True):
+        super().__init__()
+        self.in_channels = 64
+        self.conv = conv_op(3, 64, 3, 1, 1)
+        self.bn = BatchNorm2d(64)
+        self.layer1 = self._make_layer(block, 64, layers[0], 1, quirk)
+        self.layer2 = self._make_layer(block, 128, layers[1], 2, quirk)
+        self.layer3 = self._make_layer(block, 256, layers[2], 2, quirk)
+        self.layer4 = self._make_layer(block, 512, layers[3], 2, quirk)
+        from .layers import Linear
+        self.fc = Linear(512 * block.expansion, num_classes)
+
+    def _make_layer(self, block, out_ch, blocks, stride, quirk):
+        layers = []
+        downsample = (stride != 1
+                      or self.in_channels != out_ch * block.expansion)
+        layers.append(block(self.in_channels, out_ch, stride, downsample))
+        self.in_channels = out_ch * block.expansion
+        # reference quirk (resnet.hpp:172-178): `blocks` MORE blocks after the
+        # strided one (layers[i]+1 total); standard builds blocks-1 more.
+        extra = blocks if quirk else blocks - 1
+        for _ in range(extra):
+            layers.append(block(self.in_channels, out_ch))
+        return nn.Sequential(*layers)
+
+    def forward(self, x):
+        x = O.to_compute(x)
+        out = self.bn(self.conv(x), fuse_relu=True)
+        out = self.layer1(out)
+        out = self.layer2(out)
+        out = self.layer3(out)
+        out = self.layer4(out)
+        out = O.avg_pool(out, 4)
+        out = O.flatten_features(out)
+        out = self.fc(out)
+        return out if self.training else O.log_softmax(out)
+
+
+_CONFIGS = {
+    "resnet18": (BasicBlock, (2, 2, 2, 2)),
+    "resnet34": (BasicBlock, (3, 4, 6, 3)),
+    "resnet50": (BottleNeck, (3, 4, 6, 3)),
+    "resnet101": (BottleNeck, (3, 4, 23, 3)),
+    "resnet152": (BottleNeck, (3, 8, 36, 3)),
+}
+
+
+def resnet_factory(name: str, num_classes: int = 10) -> ResNet:
+    quirk = name.endswith("q")
+    base = name[:-1] if quirk else name
+    if base not in _CONFIGS:
+        raise ValueError(f"unknown resnet {name!r}")
+    block, layers = _CONFIGS[base]
+    return ResNet(block, layers, num_classes, quirk=quirk)

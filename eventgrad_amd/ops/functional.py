@@ -1,0 +1,345 @@
+"""Differentiable op layer: HIP-native on GPU (NHWC, bf16), torch on CPU.
+
+GPU activation convention: NHWC ("channels-last") bf16 tensors of shape
+[N, H, W, C] — the natural layout for MFMA implicit-GEMM convolutions on
+CDNA4 (contiguous C is both the GEMM reduction-dim vector axis and the
+coalesced memory axis). Parameters stay fp32 in standard PyTorch layouts
+(OIHW conv weights) so checkpoints / communication match the reference
+semantics (flat fp32 per-tensor messages); weights are
+cast/permuted to bf16 KRSC per forward (cheap: weights ≪ activations).
+
+CPU paths use plain fp32 PyTorch NCHW ops — they are the numerics oracle
+referenced by tests/ (SURVEY.md §4 strategy (a)).
+
+Reference op inventory being covered: SURVEY.md §2.6
+(conv2d, BatchNorm2d, ReLU, max_pool2d(2), avg_pool2d(4), Dropout/2d,
+Linear, log_softmax+nll_loss, SGD step, per-tensor L2 norm, (p+l+r)/3,
+top-k, pack/flatten, argmax-accuracy).
+"""
+
+from __future__ import annotations
+
+import torch
+import torch.nn.functional as F
+
+from .backend import native
+
+
+def _empty_f32(device):
+    return torch.empty(0, dtype=torch.float32, device=device)
+
+
+# --------------------------------------------------------------------------
+# layout / dtype glue
+# --------------------------------------------------------------------------
+
+def to_compute(x: torch.Tensor) -> torch.Tensor:
+    """NCHW fp32 input -> compute layout: NHWC bf16 on GPU, unchanged on CPU."""
+    if x.is_cuda:
+        return x.permute(0, 2, 3, 1).contiguous().to(torch.bfloat16)
+    return x
+
+
+def flatten_features(x: torch.Tensor) -> torch.Tensor:
+    """Flatten conv features to [N, C*H*W] in NCHW element order.
+
+    The reference flattens NCHW (e.g. x.view({-1, 500}), dmnist/event/
+    event.cpp:71), so the GPU NHWC path permutes back first to keep fc weight
+    element-order identical between backends.
+    """
+    if x.is_cuda:
+        return x.permute(0, 3, 1, 2).reshape(x.shape[0], -1)
+    return x.reshape(x.shape[0], -1)
+
+
+def _w_krsc(w: torch.Tensor) -> torch.Tensor:
+    """OIHW fp32 -> KRSC bf16 ([K,R,S,C]) for the fwd implicit GEMM."""
+    return w.permute(0, 2, 3, 1).contiguous().to(torch.bfloat16)
+
+
+# --------------------------------------------------------------------------
+# conv2d
+# --------------------------------------------------------------------------
+
+class _ConvNHWC(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, w, bias, stride, padding):
+        core = native()
+        wk = _w_krsc(w)
+        b = bias.detach() if bias is not None else _empty_f32(x.device)
+        y = core.conv2d_fwd(x, wk, b, stride, padding)
+        ctx.save_for_backward(x, wk)
+        ctx.stride, ctx.padding = stride, padding
+        ctx.has_bias = bias is not None
+        ctx.hw = (x.shape[1], x.shape[2])
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        core = native()
+        x, wk = ctx.saved_tensors
+        dy = dy.contiguous()
+        dx = dw = db = None
+        if ctx.needs_input_grad[0]:
+            # dgrad weight layout: [R,S,K,C] from [K,R,S,C]
+            w_rskc = wk.permute(1, 2, 0, 3).contiguous()
+            dx = core.conv2d_dgrad(dy, w_rskc, ctx.stride, ctx.padding,
+                                   ctx.hw[0], ctx.hw[1])
+        if ctx.needs_input_grad[1]:
+            dw_krsc = core.conv2d_wgrad(x, dy, wk.shape[1], wk.shape[2],
+                                        ctx.stride, ctx.padding)
+            dw = dw_krsc.permute(0, 3, 1, 2).contiguous()  # KRSC -> OIHW fp32
+        if ctx.has_bias and ctx.needs_input_grad[2]:
+            db = core.channel_sum(dy)
+        return dx, dw, db, None, None
+
+
+def conv2d(x, w, bias=None, stride=1, padding=0):
+    if x.is_cuda:
+        return _ConvNHWC.apply(x, w, bias, int(stride), int(padding))
+    return F.conv2d(x, w, bias, stride=stride, padding=padding)
+
+
+# --------------------------------------------------------------------------
+# batch norm (+ optional fused ReLU)
+# --------------------------------------------------------------------------
+
+class _BatchNormNHWC(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, gamma, beta, running_mean, running_var, training,
+                momentum, eps, fuse_relu):
+        core = native()
+        y, save_mean, save_invstd = core.bn_fwd(
+            x, gamma.detach(), beta.detach(), running_mean, running_var,
+            momentum, eps, training, fuse_relu)
+        ctx.save_for_backward(x, gamma, save_mean, save_invstd, y)
+        ctx.fuse_relu = fuse_relu
+        ctx.training = training
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        core = native()
+        x, gamma, save_mean, save_invstd, y = ctx.saved_tensors
+        dx, dgamma, dbeta = core.bn_bwd(dy.contiguous(), x, save_mean,
+                                        save_invstd, gamma.detach(), y,
+                                        ctx.fuse_relu, ctx.training)
+        return dx, dgamma, dbeta, None, None, None, None, None, None
+
+
+def batch_norm(x, gamma, beta, running_mean, running_var, training,
+               momentum=0.1, eps=1e-5, fuse_relu=False):
+    if x.is_cuda:
+        return _BatchNormNHWC.apply(x, gamma, beta, running_mean, running_var,
+                                    training, momentum, eps, fuse_relu)
+    y = F.batch_norm(x, running_mean, running_var, gamma, beta, training,
+                     momentum, eps)
+    return F.relu(y) if fuse_relu else y
+
+
+# --------------------------------------------------------------------------
+# elementwise: relu / residual add+relu / dropout
+# --------------------------------------------------------------------------
+
+class _ReLU(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x):
+        y = native().relu_fwd(x)
+        ctx.save_for_backward(y)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        (y,) = ctx.saved_tensors
+        return native().relu_bwd(dy.contiguous(), y)
+
+
+def relu(x):
+    return _ReLU.apply(x) if x.is_cuda else F.relu(x)
+
+
+class _AddReLU(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, a, b):
+        y = native().add_relu_fwd(a, b)
+        ctx.save_for_backward(y)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        (y,) = ctx.saved_tensors
+        da = native().relu_bwd(dy.contiguous(), y)
+        return da, da
+
+
+def add_relu(a, b):
+    """Residual join: relu(a + b) (resnet.hpp:46-48)."""
+    if a.is_cuda:
+        return _AddReLU.apply(a, b)
+    return F.relu(a + b)
+
+
+class _Dropout(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, p, seed, per_channel):
+        y, mask = native().dropout_fwd(x, p, seed, per_channel)
+        ctx.save_for_backward(mask)
+        ctx.p, ctx.per_channel = p, per_channel
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        (mask,) = ctx.saved_tensors
+        return (native().dropout_bwd(dy.contiguous(), mask, ctx.p,
+                                     ctx.per_channel), None, None, None)
+
+
+def _draw_seed() -> int:
+    # consume the (manually-seeded) CPU RNG so dropout is run-deterministic
+    return int(torch.randint(0, 2**31 - 1, (1,)).item())
+
+
+def dropout(x, p, training):
+    if not training or p == 0.0:
+        return x
+    if x.is_cuda:
+        return _Dropout.apply(x, p, _draw_seed(), False)
+    return F.dropout(x, p, training)
+
+
+def dropout2d(x, p, training):
+    """Channel dropout. GPU input is NHWC; mask is per (n, c)."""
+    if not training or p == 0.0:
+        return x
+    if x.is_cuda:
+        return _Dropout.apply(x, p, _draw_seed(), True)
+    return F.dropout2d(x, p, training)
+
+
+# --------------------------------------------------------------------------
+# pooling
+# --------------------------------------------------------------------------
+
+class _MaxPool2x2(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x):
+        y, idx = native().maxpool2x2_fwd(x)
+        ctx.save_for_backward(idx)
+        ctx.hw = (x.shape[1], x.shape[2])
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        (idx,) = ctx.saved_tensors
+        return native().maxpool2x2_bwd(dy.contiguous(), idx, *ctx.hw)
+
+
+def max_pool2x2(x):
+    """max_pool2d(kernel=2, stride=2) with floor semantics (event.cpp:68-70)."""
+    if x.is_cuda:
+        return _MaxPool2x2.apply(x)
+    return F.max_pool2d(x, 2)
+
+
+class _AvgPool(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, k):
+        ctx.k = k
+        ctx.hw = (x.shape[1], x.shape[2])
+        return native().avgpool_fwd(x, k)
+
+    @staticmethod
+    def backward(ctx, dy):
+        return native().avgpool_bwd(dy.contiguous(), ctx.k, *ctx.hw), None
+
+
+def avg_pool(x, k):
+    """avg_pool2d(kernel=k, stride=k) (resnet.hpp:152)."""
+    if x.is_cuda:
+        return _AvgPool.apply(x, k)
+    return F.avg_pool2d(x, k)
+
+
+# --------------------------------------------------------------------------
+# linear
+# --------------------------------------------------------------------------
+
+class _Linear(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, w, bias):
+        core = native()
+        wt = w.t().contiguous().to(torch.bfloat16)          # [K, N]
+        b = bias.detach() if bias is not None else _empty_f32(x.device)
+        y = core.gemm_bias(x, wt, b, True)
+        ctx.save_for_backward(x, w)
+        ctx.has_bias = bias is not None
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        core = native()
+        x, w = ctx.saved_tensors
+        dy = dy.contiguous()
+        dx = dw = db = None
+        e = _empty_f32(x.device)
+        if ctx.needs_input_grad[0]:
+            wb = w.contiguous().to(torch.bfloat16)           # [N, K]
+            dx = core.gemm_bias(dy, wb, e, True)
+        if ctx.needs_input_grad[1]:
+            dyt = dy.t().contiguous()                        # [N, M] bf16
+            dw = core.gemm_bias(dyt, x, e, False)            # fp32 [N, K]
+        if ctx.has_bias and ctx.needs_input_grad[2]:
+            db = core.channel_sum(dy)
+        return dx, dw, db
+
+
+def linear(x, w, bias=None):
+    if x.is_cuda:
+        return _Linear.apply(x, w, bias)
+    return F.linear(x, w, bias)
+
+
+# --------------------------------------------------------------------------
+# loss
+# --------------------------------------------------------------------------
+
+class _LogSoftmaxNLL(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, logits, target):
+        loss, logp = native().logsoftmax_nll_fwd(logits, target)
+        ctx.save_for_backward(logp, target)
+        ctx.out_dtype = logits.dtype
+        return loss
+
+    @staticmethod
+    def backward(ctx, dloss):
+        logp, target = ctx.saved_tensors
+        d = native().logsoftmax_nll_bwd(logp, target, float(dloss.item()))
+        return d.to(ctx.out_dtype), None
+
+
+def nll_of_logits(logits, target):
+    """mean nll_loss(log_softmax(logits)) (cent.cpp:119, event.cpp:291).
+
+    Note the reference models already return log_softmax and the loss applies
+    log_softmax again; log_softmax is idempotent so a single application is
+    mathematically identical — we apply it exactly once here.
+    """
+    if logits.is_cuda:
+        return _LogSoftmaxNLL.apply(logits, target)
+    return F.nll_loss(F.log_softmax(logits.float(), dim=1), target)
+
+
+def log_softmax(logits):
+    if logits.is_cuda:
+        # forward-only helper (eval path); reuse the fused fwd's logp
+        _, logp = native().logsoftmax_nll_fwd(
+            logits, torch.zeros(logits.shape[0], dtype=torch.long,
+                                device=logits.device))
+        return logp
+    return F.log_softmax(logits.float(), dim=1)
+
+
+def accuracy_count(logits, target) -> int:
+    """argmax-eq-sum accuracy accumulator (cent.cpp:147-148)."""
+    return int((logits.float().argmax(dim=1) == target).sum().item())

@@ -1,0 +1,101 @@
+"""EventGraD trigger + adaptive-threshold controller (host mirror).
+
+Semantics transcribed from dmnist/event/event.cpp:324-392 (identical in
+dcifar10/event/event.cpp:299-365 and spevent.cpp:321-426):
+
+  per pass, per tensor i (norm = ||theta_i||_2 of the CURRENT parameter,
+  i.e. the value produced by the previous optimizer step, before this
+  pass's neighbor averaging):
+
+    value_diff = |norm_i - last_sent_norm[i]|
+    iter_diff  = pass_num - last_sent_iter[i]
+    thres[i]   = thres[i] * horizon      (adaptive)   # pre-update
+               = constant                (static)
+    fire_i     = value_diff >= thres[i]  or  pass_num < initial_comm_passes
+    if fire_i:
+        slopes[i] <- shift left; slopes[i][-1] = value_diff / iter_diff
+        if adaptive: thres[i] = mean(slopes[i])
+        last_sent_norm[i] = norm_i ; last_sent_iter[i] = pass_num
+        num_events += 2                  # one per neighbor
+
+The controller state is tiny (sz <= 86 scalars per array) — on GPU the same
+update runs device-resident in the HIP trigger kernel (csrc/sgd_trigger.hip)
+and this class is used for CPU runs, unit tests, and kernel parity checks.
+"""
+
+from __future__ import annotations
+
+import numpy as np
+
+
+class TriggerController:
+    def __init__(self, sz: int, adaptive: bool, horizon: float,
+                 constant: float, sent_history: int = 2,
+                 initial_comm_passes: int = 30,
+                 always_fire: bool = False):
+        self.sz = sz
+        self.adaptive = adaptive
+        self.horizon = np.float32(horizon)
+        self.constant = np.float32(constant)
+        self.sent_history = sent_history
+        self.initial_comm_passes = initial_comm_passes
+        self.always_fire = always_fire
+
+        self.thres = np.zeros(sz, dtype=np.float32)
+        self.last_sent_norm = np.zeros(sz, dtype=np.float32)
+        self.last_sent_iter = np.zeros(sz, dtype=np.float32)
+        self.slopes = np.zeros((sz, sent_history), dtype=np.float32)
+        self.num_events = 0
+        # last evaluated values, for trace files
+        self.last_norms = np.zeros(sz, dtype=np.float32)
+        self.last_fired = np.zeros(sz, dtype=bool)
+
+    def step(self, norms: np.ndarray, pass_num: int) -> np.ndarray:
+        """Evaluate the trigger for one pass; returns fire mask (bool[sz])."""
+        norms = norms.astype(np.float32, copy=False)
+        if self.always_fire:
+            fire = np.ones(self.sz, dtype=bool)
+        else:
+            value_diff = np.abs(norms - self.last_sent_norm)
+            iter_diff = np.float32(pass_num) - self.last_sent_iter
+            if self.adaptive:
+                self.thres = self.thres * self.horizon
+            else:
+                self.thres = np.full(self.sz, self.constant, dtype=np.float32)
+            fire = (value_diff >= self.thres) | (pass_num <
+                                                 self.initial_comm_passes)
+        if fire.any():
+            value_diff = np.abs(norms - self.last_sent_norm)
+            iter_diff = np.maximum(
+                np.float32(pass_num) - self.last_sent_iter, 1.0)
+            new_slope = (value_diff / iter_diff).astype(np.float32)
+            shifted = np.roll(self.slopes, -1, axis=1)
+            shifted[:, -1] = new_slope
+            self.slopes = np.where(fire[:, None], shifted, self.slopes)
+            slope_avg = self.slopes.mean(axis=1, dtype=np.float32)
+            if self.adaptive and not self.always_fire:
+                self.thres = np.where(fire, slope_avg, self.thres)
+            self.last_sent_norm = np.where(fire, norms, self.last_sent_norm)
+            self.last_sent_iter = np.where(
+                fire, np.float32(pass_num), self.last_sent_iter)
+        self.num_events += 2 * int(fire.sum())
+        self.last_norms = norms.copy()
+        self.last_fired = fire.copy()
+        return fire
+
+    # -- checkpointing -----------------------------------------------------
+    def state_dict(self) -> dict:
+        return {
+            "thres": self.thres.copy(),
+            "last_sent_norm": self.last_sent_norm.copy(),
+            "last_sent_iter": self.last_sent_iter.copy(),
+            "slopes": self.slopes.copy(),
+            "num_events": self.num_events,
+        }
+
+    def load_state_dict(self, d: dict) -> None:
+        self.thres = d["thres"].astype(np.float32).copy()
+        self.last_sent_norm = d["last_sent_norm"].astype(np.float32).copy()
+        self.last_sent_iter = d["last_sent_iter"].astype(np.float32).copy()
+        self.slopes = d["slopes"].astype(np.float32).copy()
+        self.num_events = int(d["num_events"])

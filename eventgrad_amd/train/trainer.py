@@ -1,0 +1,152 @@
+"""Training loop for all modes (the reference's per-binary main()s unified).
+
+Skeleton: epoch loop -> batch loop -> [comm begin_pass] ->
+zero_grad -> forward -> loss -> backward -> [comm after_backward: allreduce
+or wait+unpack+average] -> fused SGD step (+ per-tensor norms) -> accuracy;
+then final consensus allreduce + event-count reduction + rank-0 test eval.
+"""
+
+from __future__ import annotations
+
+import time
+
+import torch
+
+from ..config import RunConfig
+from ..data import build_dataset
+from ..data.loader import ShardLoader
+from ..data.sampler import (DistributedRandomSampler,
+                            DistributedSequentialSampler)
+from ..models import build_model
+from ..ops import functional as O
+from ..parallel import FlatParamSpace, build_engine, init_distributed
+from .checkpoint import load_checkpoint, save_checkpoint
+from .metrics import RunMetrics
+from .trace import Tracer
+
+
+class Trainer:
+    def __init__(self, cfg: RunConfig):
+        self.cfg = cfg
+        self.rank, self.world, self.device = init_distributed(cfg.device)
+        torch.manual_seed(cfg.seed)  # ref: torch::manual_seed(0) everywhere
+
+        self.model = build_model(cfg.model, cfg.data.num_classes)
+        self.model.to(self.device)
+        self.space = FlatParamSpace(self.model, self.device)
+
+        self.tracer = Tracer(self.rank, cfg.trace_dir) if cfg.trace else None
+        self.engine = build_engine(self.space, cfg, self.rank, self.world,
+                                   self.device, self.tracer)
+
+        self.train_set = build_dataset(cfg.data, train=True)
+        self.test_set = build_dataset(cfg.data, train=False)
+        bs = cfg.data.batch_size
+        if cfg.data.global_batch:
+            bs = max(1, cfg.data.global_batch // self.world)
+        sampler_cls = (DistributedRandomSampler if cfg.data.shuffle
+                       else DistributedSequentialSampler)
+        if cfg.data.shuffle:
+            sampler = sampler_cls(len(self.train_set), self.world, self.rank,
+                                  seed=cfg.seed)
+        else:
+            sampler = sampler_cls(len(self.train_set), self.world, self.rank)
+        self.loader = ShardLoader(self.train_set, sampler, bs,
+                                  augment=cfg.data.augment, seed=cfg.seed)
+        self.metrics = RunMetrics(world=self.world,
+                                  num_tensors=self.space.sz)
+        self.pass_num = 0
+        self.start_epoch = 1
+        if cfg.resume and cfg.checkpoint_path:
+            st = load_checkpoint(cfg.checkpoint_path, self.space, self.engine,
+                                 self.device)
+            self.pass_num = st["pass_num"]
+            self.start_epoch = st["epoch"] + 1
+
+    # ------------------------------------------------------------------
+    def train(self) -> RunMetrics:
+        cfg = self.cfg
+        self.model.train()
+        t0 = time.perf_counter()
+        for epoch in range(self.start_epoch, cfg.epochs + 1):
+            correct = seen = 0
+            last_loss = 0.0
+            for x, y in self.loader.epoch(epoch):
+                x = x.to(self.device, non_blocking=True)
+                y = y.to(self.device, non_blocking=True)
+                self.pass_num += 1
+                self.engine.begin_pass(self.pass_num)
+                self.space.zero_grad()
+                logits = self.model(x)
+                loss = O.nll_of_logits(logits, y)
+                loss.backward()
+                self.engine.after_backward()
+                self.engine.step()
+                correct += O.accuracy_count(logits.detach(), y)
+                seen += x.shape[0]
+                last_loss = float(loss.detach())
+            acc = 100.0 * correct / max(seen, 1)
+            self.metrics.epoch_train_acc.append(round(acc, 4))
+            self.metrics.final_train_loss = last_loss
+            self.metrics.samples_seen += seen
+            if self.rank == 0:
+                print(f"{epoch}, {acc}", flush=True)  # decent.cpp:255 format
+            if self.tracer:
+                self.tracer.train_line(epoch, acc, last_loss)
+            if (cfg.checkpoint_path and cfg.checkpoint_every_epochs
+                    and epoch % cfg.checkpoint_every_epochs == 0):
+                save_checkpoint(cfg.checkpoint_path, cfg, epoch,
+                                self.pass_num, self.space, self.engine)
+        if self.device.type == "cuda":
+            torch.cuda.synchronize()
+        self.metrics.train_time_s = time.perf_counter() - t0
+        self.metrics.total_passes = self.pass_num
+
+        # closing consensus + event aggregation (event.cpp:517-532)
+        if cfg.final_consensus:
+            stats = self.engine.finalize()
+            self.metrics.num_events_total = stats["num_events_total"]
+        else:
+            self.metrics.num_events_total = self.engine.num_events * self.world
+        if self.rank == 0:
+            print(f"Training time - {self.metrics.train_time_s}", flush=True)
+            print(f"Total number of events - "
+                  f"{self.metrics.num_events_total}", flush=True)
+
+        if cfg.checkpoint_path:
+            save_checkpoint(cfg.checkpoint_path, cfg, cfg.epochs,
+                            self.pass_num, self.space, self.engine)
+        if cfg.eval_at_end and self.rank == 0:
+            self.evaluate()
+        if self.tracer:
+            self.tracer.close()
+        return self.metrics
+
+    # ------------------------------------------------------------------
+    @torch.no_grad()
+    def evaluate(self, batch_size: int = 100) -> float:
+        """Rank-0 held-out eval (cent.cpp:166-214, event.cpp:535-586)."""
+        self.model.eval()
+        n = len(self.test_set)
+        correct = 0
+        total_loss = 0.0
+        import numpy as np
+        for s in range(0, n, batch_size):
+            idx = np.arange(s, min(s + batch_size, n))
+            x, y = self.test_set.batch(idx)
+            x, y = x.to(self.device), y.to(self.device)
+            logp = self.model(x)  # eval mode returns log-probs
+            total_loss += float(
+                torch.nn.functional.nll_loss(logp.float(), y,
+                                             reduction="sum"))
+            correct += O.accuracy_count(logp, y)
+        acc = 100.0 * correct / n
+        self.metrics.test_accuracy = acc
+        self.metrics.test_loss = total_loss / n
+        print(f"Test Accuracy - {acc}", flush=True)
+        self.model.train()
+        return acc
+
+
+def train_from_config(cfg: RunConfig) -> RunMetrics:
+    return Trainer(cfg).train()
