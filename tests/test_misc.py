@@ -1,0 +1,111 @@
+"""Config/CLI/data/metrics/trace unit tests."""
+
+import os
+
+import numpy as np
+import torch
+
+from eventgrad_amd.config import RunConfig, preset
+from eventgrad_amd.data import (Cifar10Dataset, DistributedRandomSampler,
+                                DistributedSequentialSampler, MnistDataset,
+                                SyntheticImages)
+from eventgrad_amd.data.loader import ShardLoader
+from eventgrad_amd.data.transforms import augment_batch
+from eventgrad_amd.train.cli import build_parser, config_from_args
+from eventgrad_amd.train.metrics import RunMetrics
+
+
+def test_presets_roundtrip():
+    for name in ("dmnist-cent", "dmnist-event", "dcifar10-event",
+                 "dcifar10-spevent"):
+        cfg = preset(name)
+        s = cfg.to_json()
+        cfg2 = RunConfig.from_json(s)
+        assert cfg2 == cfg
+
+
+def test_cli_maps_reference_args():
+    p = build_parser()
+    args = p.parse_args(["--preset", "dcifar10-event", "--thres-constant",
+                         "0.002", "--epochs", "3", "--topk-percent", "2.5"])
+    cfg = config_from_args(args)
+    assert cfg.mode == "event" and cfg.model == "resnet18q"
+    assert not cfg.trigger.adaptive and cfg.trigger.constant == 0.002
+    assert cfg.epochs == 3 and cfg.topk_percent == 2.5
+    assert cfg.data.global_batch == 256
+    assert cfg.optim.momentum == 0.9
+
+
+def test_synthetic_deterministic_and_learnable():
+    ds = SyntheticImages((3, 32, 32), 100, noise=0.3)
+    x1, y1 = ds[7]
+    x2, y2 = ds[7]
+    assert torch.equal(x1, x2) and y1 == y2
+    # same-class samples closer than cross-class on average
+    xa, _ = ds[0]
+    xb, _ = ds[10]  # same class (10 % 10 == 0)
+    xc, _ = ds[1]
+    assert (xa - xb).norm() < (xa - xc).norm()
+
+
+def test_samplers_shard_disjoint():
+    seq = [DistributedSequentialSampler(100, 4, r) for r in range(4)]
+    allidx = np.concatenate([s.epoch_indices(0) for s in seq])
+    assert len(set(allidx.tolist())) == 100
+    rnd = [DistributedRandomSampler(100, 4, r, seed=0) for r in range(4)]
+    allidx = np.concatenate([s.epoch_indices(3) for s in rnd])
+    assert len(set(allidx.tolist())) == 100
+    assert not np.array_equal(rnd[0].epoch_indices(0),
+                              rnd[0].epoch_indices(1))
+
+
+def test_loader_batches_and_augment():
+    ds = SyntheticImages((3, 32, 32), 64)
+    s = DistributedSequentialSampler(64, 2, 0)
+    ld = ShardLoader(ds, s, 8, augment=True, seed=0)
+    batches = list(ld.epoch(1))
+    assert len(batches) == 4
+    x, y = batches[0]
+    assert x.shape == (8, 3, 32, 32) and y.shape == (8,)
+
+
+def test_augment_shapes_and_determinism():
+    x = torch.randn(4, 3, 32, 32)
+    g1 = torch.Generator().manual_seed(5)
+    g2 = torch.Generator().manual_seed(5)
+    a = augment_batch(x, generator=g1)
+    b = augment_batch(x, generator=g2)
+    assert a.shape == x.shape and torch.equal(a, b)
+
+
+def test_messages_saved_math():
+    m = RunMetrics(world=4, num_tensors=8, total_passes=100,
+                   num_events_total=4 * 2 * 8 * 30)  # only warmup fired
+    assert abs(m.messages_saved_pct - 70.0) < 1e-9
+
+
+def test_mnist_idx_parser(tmp_path):
+    import struct
+    imgs = np.random.randint(0, 255, (10, 28, 28), dtype=np.uint8)
+    lbls = np.random.randint(0, 10, (10,), dtype=np.uint8)
+    with open(os.path.join(tmp_path, "train-images-idx3-ubyte"), "wb") as f:
+        f.write(struct.pack(">BBBBIII", 0, 0, 8, 3, 10, 28, 28))
+        f.write(imgs.tobytes())
+    with open(os.path.join(tmp_path, "train-labels-idx1-ubyte"), "wb") as f:
+        f.write(struct.pack(">BBBBI", 0, 0, 8, 1, 10))
+        f.write(lbls.tobytes())
+    ds = MnistDataset(str(tmp_path), train=True)
+    assert len(ds) == 10
+    x, y = ds[3]
+    assert x.shape == (1, 28, 28) and y == int(lbls[3])
+
+
+def test_cifar_binary_parser(tmp_path):
+    raw = np.random.randint(0, 255, (5, 3073), dtype=np.uint8)
+    raw[:, 0] = np.arange(5) % 10
+    for i in range(1, 6):
+        raw.tofile(os.path.join(tmp_path, f"data_batch_{i}.bin"))
+    ds = Cifar10Dataset(str(tmp_path), train=True)
+    assert len(ds) == 25
+    x, y = ds[0]
+    assert x.shape == (3, 32, 32) and 0 <= y < 10
