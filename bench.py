@@ -1,0 +1,139 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: CIFAR-10 quirk-ResNet EventGraD training step.
+
+Measures whole-job samples/sec for the BASELINE.json headline config
+("CIFAR-10 ResNet EventGraD (dcifar10/event) ring"): the reference's coded
+model ResNet<BasicBlock>({2,2,2,2}) with its make_layer quirk — 17,444,682
+params / 86 tensors — SGD lr 1e-2 momentum 0.9, adaptive
+event trigger (horizon 1.01, 30 warmup passes), synthetic CIFAR-shaped data
+(32x32x3, random-init weights; no network in this environment). Per-GPU
+batch is fixed at 256 (the reference's global batch at 1 rank), so scaling
+is WEAK; the driver computes scaling efficiency from per-N runs.
+
+Contract: one rank per GPU via torch.distributed.run; rank 0 prints ONE
+JSON line. Timed region = K full training steps (data->device, trigger +
+ring gossip, forward, loss, backward, neighbor averaging, fused SGD step)
+bracketed by barrier + torch.cuda.synchronize on both sides; value uses the
+MAX elapsed over ranks.
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import time
+
+import torch
+
+from eventgrad_amd.config import preset
+from eventgrad_amd.models import build_model
+from eventgrad_amd.ops import functional as O
+from eventgrad_amd.parallel import FlatParamSpace, build_engine, init_distributed
+
+
+def main() -> int:
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=30)
+    ap.add_argument("--warmup", type=int, default=10)
+    ap.add_argument("--batch", type=int, default=256, help="per-GPU batch")
+    ap.add_argument("--mode", default="event",
+                    choices=["event", "decent", "cent", "spevent", "serial"])
+    ap.add_argument("--model", default="resnet18q")
+    args = ap.parse_args()
+
+    cfg = preset("dcifar10-event")
+    cfg.mode = args.mode
+    cfg.model = args.model
+    cfg.data.global_batch = None
+    cfg.data.batch_size = args.batch
+
+    rank, world, device = init_distributed("auto")
+    if device.type != "cuda":
+        raise SystemExit("bench.py requires a GPU")
+    torch.manual_seed(cfg.seed)
+
+    model = build_model(cfg.model).to(device)
+    model.train()
+    space = FlatParamSpace(model, device)
+    engine = build_engine(space, cfg, rank, world, device)
+
+    # synthetic data, resident batches (data=synthetic per BASELINE contract)
+    g = torch.Generator(device="cpu").manual_seed(1234 + rank)
+    n_resident = 8
+    xs = [torch.randn(args.batch, 3, 32, 32, generator=g).to(device)
+          for _ in range(n_resident)]
+    ys = [torch.randint(0, 10, (args.batch,), generator=g).to(device)
+          for _ in range(n_resident)]
+
+    pass_num = 0
+
+    def step():
+        nonlocal pass_num
+        pass_num += 1
+        x, y = xs[pass_num % n_resident], ys[pass_num % n_resident]
+        engine.begin_pass(pass_num)
+        space.zero_grad()
+        logits = model(x)
+        loss = O.nll_of_logits(logits, y)
+        loss.backward()
+        engine.after_backward()
+        engine.step()
+        return loss
+
+    def barrier_sync():
+        if world > 1:
+            torch.distributed.barrier()
+        torch.cuda.synchronize()
+
+    for _ in range(args.warmup):
+        step()
+    barrier_sync()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        loss = step()
+    barrier_sync()
+    elapsed = time.perf_counter() - t0
+
+    if world > 1:
+        t = torch.tensor([elapsed], device=device)
+        torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    if rank == 0:
+        samples = float(world * args.batch * args.steps)
+        out = {
+            "metric": "samples_per_sec",
+            "value": round(samples / elapsed, 2),
+            "unit": "samples/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(1000.0 * elapsed / args.steps, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16",
+            "data": "synthetic",
+            "config": {
+                "model": "resnet18-quirk (ref ResNet<BasicBlock>{2,2,2,2}, "
+                         "17444682 params / 86 tensors)",
+                "mode": args.mode,
+                "global_batch": world * args.batch,
+                "per_gpu_batch": args.batch,
+                "seq_len": None,
+                "parallelism": f"eventgrad-ring dp{world}",
+                "optimizer": "sgd lr=1e-2 momentum=0.9",
+                "trigger": "adaptive horizon=1.01 warmup=30",
+                "final_loss": round(float(loss.item()), 4),
+            },
+        }
+        print(json.dumps(out), flush=True)
+    if world > 1:
+        torch.distributed.destroy_process_group()
+    return 0
+
+
+if __name__ == "__main__":
+    raise SystemExit(main())

@@ -1,0 +1,83 @@
+// pybind bindings for eventgrad_amd._core (gfx950 HIP kernels).
+#include <torch/extension.h>
+
+#include <vector>
+
+namespace eg {
+// engine.hip
+torch::Tensor sgd_step_norm(torch::Tensor, torch::Tensor, torch::Tensor,
+                            torch::Tensor, torch::Tensor, double, double,
+                            double);
+torch::Tensor seg_sqnorms(torch::Tensor, torch::Tensor, torch::Tensor);
+void avg3(torch::Tensor, torch::Tensor, torch::Tensor);
+torch::Tensor gather_segments(torch::Tensor, torch::Tensor, torch::Tensor,
+                              long);
+void scatter_segments(torch::Tensor, torch::Tensor, torch::Tensor,
+                      torch::Tensor);
+torch::Tensor trigger_update(torch::Tensor, torch::Tensor, torch::Tensor,
+                             torch::Tensor, torch::Tensor, torch::Tensor,
+                             long, bool, double, double, long, bool);
+// elementwise.hip
+torch::Tensor relu_fwd(torch::Tensor);
+torch::Tensor relu_bwd(torch::Tensor, torch::Tensor);
+torch::Tensor add_relu_fwd(torch::Tensor, torch::Tensor);
+std::vector<torch::Tensor> dropout_fwd(torch::Tensor, double, long, bool);
+torch::Tensor dropout_bwd(torch::Tensor, torch::Tensor, double, bool);
+torch::Tensor channel_sum(torch::Tensor);
+// gemm.hip
+torch::Tensor gemm_bias(torch::Tensor, torch::Tensor, torch::Tensor, bool);
+// conv.hip
+torch::Tensor conv2d_fwd(torch::Tensor, torch::Tensor, torch::Tensor, long,
+                         long);
+torch::Tensor conv2d_dgrad(torch::Tensor, torch::Tensor, long, long, long,
+                           long);
+torch::Tensor conv2d_wgrad(torch::Tensor, torch::Tensor, long, long, long,
+                           long);
+// bn.hip
+std::vector<torch::Tensor> bn_fwd(torch::Tensor, torch::Tensor, torch::Tensor,
+                                  torch::Tensor, torch::Tensor, double,
+                                  double, bool, bool);
+std::vector<torch::Tensor> bn_bwd(torch::Tensor, torch::Tensor, torch::Tensor,
+                                  torch::Tensor, torch::Tensor, torch::Tensor,
+                                  bool, bool);
+// pool.hip
+std::vector<torch::Tensor> maxpool2x2_fwd(torch::Tensor);
+torch::Tensor maxpool2x2_bwd(torch::Tensor, torch::Tensor, long, long);
+torch::Tensor avgpool_fwd(torch::Tensor, long);
+torch::Tensor avgpool_bwd(torch::Tensor, long, long, long);
+// loss.hip
+std::vector<torch::Tensor> logsoftmax_nll_fwd(torch::Tensor, torch::Tensor);
+torch::Tensor logsoftmax_nll_bwd(torch::Tensor, torch::Tensor, double);
+// topk.hip
+std::vector<torch::Tensor> topk_absdiff(torch::Tensor, torch::Tensor, long);
+void scatter_update(torch::Tensor, torch::Tensor, torch::Tensor);
+}  // namespace eg
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("sgd_step_norm", &eg::sgd_step_norm);
+  m.def("seg_sqnorms", &eg::seg_sqnorms);
+  m.def("avg3", &eg::avg3);
+  m.def("gather_segments", &eg::gather_segments);
+  m.def("scatter_segments", &eg::scatter_segments);
+  m.def("trigger_update", &eg::trigger_update);
+  m.def("relu_fwd", &eg::relu_fwd);
+  m.def("relu_bwd", &eg::relu_bwd);
+  m.def("add_relu_fwd", &eg::add_relu_fwd);
+  m.def("dropout_fwd", &eg::dropout_fwd);
+  m.def("dropout_bwd", &eg::dropout_bwd);
+  m.def("channel_sum", &eg::channel_sum);
+  m.def("gemm_bias", &eg::gemm_bias);
+  m.def("conv2d_fwd", &eg::conv2d_fwd);
+  m.def("conv2d_dgrad", &eg::conv2d_dgrad);
+  m.def("conv2d_wgrad", &eg::conv2d_wgrad);
+  m.def("bn_fwd", &eg::bn_fwd);
+  m.def("bn_bwd", &eg::bn_bwd);
+  m.def("maxpool2x2_fwd", &eg::maxpool2x2_fwd);
+  m.def("maxpool2x2_bwd", &eg::maxpool2x2_bwd);
+  m.def("avgpool_fwd", &eg::avgpool_fwd);
+  m.def("avgpool_bwd", &eg::avgpool_bwd);
+  m.def("logsoftmax_nll_fwd", &eg::logsoftmax_nll_fwd);
+  m.def("logsoftmax_nll_bwd", &eg::logsoftmax_nll_bwd);
+  m.def("topk_absdiff", &eg::topk_absdiff);
+  m.def("scatter_update", &eg::scatter_update);
+}

@@ -81,9 +81,9 @@ class _ConvNHWC(torch.autograd.Function):
         dy = dy.contiguous()
         dx = dw = db = None
         if ctx.needs_input_grad[0]:
-            # dgrad weight layout: [R,S,K,C] from [K,R,S,C]
-            w_rskc = wk.permute(1, 2, 0, 3).contiguous()
-            dx = core.conv2d_dgrad(dy, w_rskc, ctx.stride, ctx.padding,
+            # dgrad weight layout: [C,R,S,K] from [K,R,S,C]
+            w_crsk = wk.permute(3, 1, 2, 0).contiguous()
+            dx = core.conv2d_dgrad(dy, w_crsk, ctx.stride, ctx.padding,
                                    ctx.hw[0], ctx.hw[1])
         if ctx.needs_input_grad[1]:
             dw_krsc = core.conv2d_wgrad(x, dy, wk.shape[1], wk.shape[2],
@@ -265,29 +265,33 @@ def avg_pool(x, k):
 # --------------------------------------------------------------------------
 
 class _Linear(torch.autograd.Function):
+    """GEMM is NT form: gemm_bias(A[M,K], B[N,K]) = A @ B.T — torch Linear
+    weight [N,K] feeds the forward without any transpose."""
+
     @staticmethod
     def forward(ctx, x, w, bias):
         core = native()
-        wt = w.t().contiguous().to(torch.bfloat16)          # [K, N]
+        wb = w.detach().to(torch.bfloat16)                   # [N, K]
         b = bias.detach() if bias is not None else _empty_f32(x.device)
-        y = core.gemm_bias(x, wt, b, True)
-        ctx.save_for_backward(x, w)
+        y = core.gemm_bias(x, wb, b, True)
+        ctx.save_for_backward(x, wb)
         ctx.has_bias = bias is not None
         return y
 
     @staticmethod
     def backward(ctx, dy):
         core = native()
-        x, w = ctx.saved_tensors
+        x, wb = ctx.saved_tensors
         dy = dy.contiguous()
         dx = dw = db = None
         e = _empty_f32(x.device)
         if ctx.needs_input_grad[0]:
-            wb = w.contiguous().to(torch.bfloat16)           # [N, K]
-            dx = core.gemm_bias(dy, wb, e, True)
+            wt = wb.t().contiguous()                         # [K, N]
+            dx = core.gemm_bias(dy, wt, e, True)
         if ctx.needs_input_grad[1]:
-            dyt = dy.t().contiguous()                        # [N, M] bf16
-            dw = core.gemm_bias(dyt, x, e, False)            # fp32 [N, K]
+            # dw[N,K] = dy^T[N,M] @ (x^T[K,M])^T, fp32 out
+            dw = core.gemm_bias(dy.t().contiguous(),
+                                x.t().contiguous(), e, False)
         if ctx.has_bias and ctx.needs_input_grad[2]:
             db = core.channel_sum(dy)
         return dx, dw, db

@@ -104,9 +104,9 @@ class _GpuK:
         src = torch.tensor([space.starts[i] for i in fired_idx],
                            dtype=torch.int64, device=dev)
         sizes = [space.numels[i] for i in fired_idx]
-        offs = torch.tensor(np.concatenate([[0], np.cumsum(sizes)]),
-                            dtype=torch.int64, device=dev)
-        return native().gather_segments(buf, src, offs)
+        cum = np.concatenate([[0], np.cumsum(sizes)])
+        offs = torch.tensor(cum, dtype=torch.int64, device=dev)
+        return native().gather_segments(buf, src, offs, int(cum[-1]))
 
     @staticmethod
     def unpack(space, payload, fired_idx, inbox) -> None:
@@ -309,16 +309,17 @@ class SparseGossipEngine(GossipEngine):
             seg = self.space.seg(self.space.param, i)
             pseg = self.space.seg(self.prev, i)
             if self.device.type == "cuda":
+                # kernel also updates prev[idx] = vals (spevent.cpp:407-413)
                 vals, idx = native().topk_absdiff(seg, pseg, self.k[i])
             else:
                 diff = (seg - pseg).abs()
                 _, idx = torch.topk(diff, self.k[i], sorted=True)
-                idx = idx.to(torch.int64)
                 vals = seg[idx]
-            # update prev at the sent indices only (spevent.cpp:407-413)
-            pseg[idx.long()] = vals
+                # update prev at the sent indices only (spevent.cpp:407-413)
+                pseg[idx] = vals
+                idx = idx.to(torch.int32)
             chunks.append(vals.to(torch.float32))
-            chunks.append(idx.to(torch.int32).view(torch.float32))
+            chunks.append(idx.view(torch.float32))
         payload = torch.cat(chunks)
         return payload, payload.clone()
 
@@ -331,8 +332,12 @@ class SparseGossipEngine(GossipEngine):
         for i in fired_idx:
             k = self.k[i]
             vals = payload[off:off + k]
-            idx = payload[off + k:off + 2 * k].view(torch.int32).long()
-            self.space.seg(replica, i)[idx] = vals
+            idx = payload[off + k:off + 2 * k].view(torch.int32)
+            seg = self.space.seg(replica, i)
+            if self.device.type == "cuda":
+                native().scatter_update(seg, vals, idx)
+            else:
+                seg[idx.long()] = vals
             off += 2 * k
 
 

@@ -1,0 +1,341 @@
+"""GPU numerics: every HIP kernel vs a plain PyTorch fp32 reference.
+
+bf16-compute kernels are compared against fp32 torch with inputs
+pre-quantized to bf16, using tolerances sized for bf16 rounding
+(~8 mantissa bits -> relative ~1e-2 after short reductions).
+"""
+
+import numpy as np
+import pytest
+import torch
+import torch.nn.functional as F
+
+pytestmark = pytest.mark.gpu
+
+DEV = "cuda"
+
+
+def core():
+    from eventgrad_amd.ops.backend import native
+    return native()
+
+
+def bq(x):
+    """quantize to bf16 then back to fp32 (the oracle's input)."""
+    return x.to(torch.bfloat16).float()
+
+
+def rel_err(a, b):
+    return ((a - b).norm() / (b.norm() + 1e-12)).item()
+
+
+# ---------------------------------------------------------------- gemm ----
+
+@pytest.mark.parametrize("M,N,K,bias,out_bf16", [
+    (37, 53, 100, False, False),
+    (64, 64, 64, True, True),
+    (256, 10, 500, True, True),
+    (128, 784, 256, False, False),
+    (1, 10, 50, True, True),
+])
+def test_gemm_bias(M, N, K, bias, out_bf16):
+    torch.manual_seed(0)
+    A = torch.randn(M, K, device=DEV).to(torch.bfloat16)
+    B = torch.randn(N, K, device=DEV).to(torch.bfloat16)
+    b = torch.randn(N, device=DEV) if bias else torch.empty(0, device=DEV)
+    C = core().gemm_bias(A, B, b, out_bf16)
+    ref = bq(A.float()).cpu() @ bq(B.float()).cpu().t()
+    if bias:
+        ref = ref + b.cpu()
+    assert rel_err(C.float().cpu(), ref) < 2e-2
+
+
+# ---------------------------------------------------------------- conv ----
+
+CONV_CASES = [
+    # N, H, W, C, K, R, stride, pad, bias
+    (2, 16, 16, 8, 16, 3, 1, 1, False),
+    (2, 32, 32, 3, 64, 3, 1, 1, False),    # stem (C=3 slow path)
+    (2, 16, 16, 64, 128, 3, 2, 1, False),  # strided stage transition
+    (2, 16, 16, 64, 128, 1, 2, 0, False),  # 1x1 downsample
+    (2, 28, 28, 1, 10, 3, 1, 0, True),     # CNN-2 conv1 (bias, no pad)
+    (2, 13, 13, 10, 20, 3, 1, 0, True),    # CNN-2 conv2
+    (2, 8, 8, 512, 512, 3, 1, 1, False),   # stage-4
+]
+
+
+@pytest.mark.parametrize("case", CONV_CASES)
+def test_conv2d_autograd(case):
+    N, H, W, C, K, R, stride, pad, bias = case
+    torch.manual_seed(1)
+    from eventgrad_amd.ops import functional as O
+
+    x_f = torch.randn(N, C, H, W)
+    w_f = torch.randn(K, C, R, R) * (1.0 / (R * np.sqrt(C)))
+    b_f = torch.randn(K) if bias else None
+
+    # GPU path (NHWC bf16)
+    x_g = bq(x_f).to(DEV).permute(0, 2, 3, 1).contiguous().to(torch.bfloat16)
+    x_g.requires_grad_(True)
+    w_g = w_f.clone().to(DEV).requires_grad_(True)
+    b_g = b_f.clone().to(DEV).requires_grad_(True) if bias else None
+    y_g = O.conv2d(x_g, w_g, b_g, stride, pad)
+
+    # CPU fp32 oracle on bf16-quantized inputs
+    x_c = bq(x_f).requires_grad_(True)
+    w_c = bq(w_f).requires_grad_(True)
+    b_c = bq(b_f).requires_grad_(True) if bias else None
+    y_c = F.conv2d(x_c, w_c, b_c, stride=stride, padding=pad)
+
+    y_g_nchw = y_g.float().permute(0, 3, 1, 2).cpu()
+    assert rel_err(y_g_nchw, y_c.detach()) < 3e-2, "fwd"
+
+    dy = torch.randn_like(y_c)
+    y_c.backward(dy)
+    dy_g = bq(dy).to(DEV).permute(0, 2, 3, 1).contiguous().to(torch.bfloat16)
+    y_g.backward(dy_g)
+
+    assert rel_err(w_g.grad.cpu(), w_c.grad) < 3e-2, "wgrad"
+    dx_g = x_g.grad.float().permute(0, 3, 1, 2).cpu()
+    assert rel_err(dx_g, x_c.grad) < 3e-2, "dgrad"
+    if bias:
+        assert rel_err(b_g.grad.cpu(), b_c.grad) < 2e-2, "bias grad"
+
+
+# ------------------------------------------------------------------ bn ----
+
+@pytest.mark.parametrize("relu", [False, True])
+def test_bn_autograd(relu):
+    torch.manual_seed(2)
+    from eventgrad_amd.ops import functional as O
+    N, H, W, C = 4, 8, 8, 32
+    x_f = torch.randn(N, C, H, W) * 2 + 1
+
+    x_g = bq(x_f).to(DEV).permute(0, 2, 3, 1).contiguous().to(torch.bfloat16)
+    x_g.requires_grad_(True)
+    g_g = torch.rand(C, device=DEV).requires_grad_(True)
+    b_g = torch.randn(C, device=DEV).requires_grad_(True)
+    rm_g = torch.zeros(C, device=DEV)
+    rv_g = torch.ones(C, device=DEV)
+    y_g = O.batch_norm(x_g, g_g, b_g, rm_g, rv_g, True, 0.1, 1e-5, relu)
+
+    x_c = bq(x_f).requires_grad_(True)
+    g_c = g_g.detach().cpu().requires_grad_(True)
+    b_c = b_g.detach().cpu().requires_grad_(True)
+    rm_c = torch.zeros(C)
+    rv_c = torch.ones(C)
+    y_c = F.batch_norm(x_c, rm_c, rv_c, g_c, b_c, True, 0.1, 1e-5)
+    if relu:
+        y_c = F.relu(y_c)
+
+    assert rel_err(y_g.float().permute(0, 3, 1, 2).cpu(), y_c.detach()) < 3e-2
+    assert rel_err(rm_g.cpu(), rm_c) < 2e-2
+    assert rel_err(rv_g.cpu(), rv_c) < 2e-2
+
+    dy = torch.randn_like(y_c)
+    y_c.backward(dy)
+    y_g.backward(bq(dy).to(DEV).permute(0, 2, 3, 1).contiguous()
+                 .to(torch.bfloat16))
+    assert rel_err(g_g.grad.cpu(), g_c.grad) < 3e-2
+    assert rel_err(b_g.grad.cpu(), b_c.grad) < 3e-2
+    dx = x_g.grad.float().permute(0, 3, 1, 2).cpu()
+    assert rel_err(dx, x_c.grad) < 4e-2
+
+
+# ----------------------------------------------------- elementwise/pool ----
+
+def test_relu_and_add_relu():
+    from eventgrad_amd.ops import functional as O
+    torch.manual_seed(3)
+    a = torch.randn(3, 4, 4, 24, device=DEV).to(torch.bfloat16)
+    b = torch.randn_like(a)
+    a.requires_grad_(True)
+    b.requires_grad_(True)
+    y = O.add_relu(a, b)
+    ref = F.relu(a.detach().float() + b.detach().float())
+    assert rel_err(y.float().cpu(), ref.cpu()) < 2e-2
+    dy = torch.randn_like(a)
+    y.backward(dy)
+    mask = (ref > 0).float()
+    assert rel_err(a.grad.float().cpu(), (dy.float() * mask).cpu()) < 2e-2
+
+    x = torch.randn(1000, device=DEV).to(torch.bfloat16).requires_grad_(True)
+    z = O.relu(x)
+    assert torch.equal(z.float().cpu(),
+                       F.relu(x.detach().float()).cpu())
+
+
+def test_pools():
+    from eventgrad_amd.ops import functional as O
+    torch.manual_seed(4)
+    x_f = torch.randn(2, 8, 12, 12)
+    xg = bq(x_f).to(DEV).permute(0, 2, 3, 1).contiguous().to(torch.bfloat16)
+    xg.requires_grad_(True)
+    y = O.max_pool2x2(xg)
+    xc = bq(x_f).requires_grad_(True)
+    yc = F.max_pool2d(xc, 2)
+    assert rel_err(y.float().permute(0, 3, 1, 2).cpu(), yc.detach()) < 1e-2
+    dy = torch.randn_like(yc)
+    yc.backward(dy)
+    y.backward(bq(dy).to(DEV).permute(0, 2, 3, 1).contiguous()
+               .to(torch.bfloat16))
+    assert rel_err(xg.grad.float().permute(0, 3, 1, 2).cpu(), xc.grad) < 2e-2
+
+    x2 = torch.randn(2, 8, 8, 16)
+    xg2 = bq(x2).to(DEV).permute(0, 2, 3, 1).contiguous().to(torch.bfloat16)
+    xg2.requires_grad_(True)
+    y2 = O.avg_pool(xg2, 4)
+    xc2 = bq(x2).requires_grad_(True)
+    yc2 = F.avg_pool2d(xc2, 4)
+    assert rel_err(y2.float().permute(0, 3, 1, 2).cpu(), yc2.detach()) < 2e-2
+    dy2 = torch.randn_like(yc2)
+    yc2.backward(dy2)
+    y2.backward(bq(dy2).to(DEV).permute(0, 2, 3, 1).contiguous()
+                .to(torch.bfloat16))
+    assert rel_err(xg2.grad.float().permute(0, 3, 1, 2).cpu(), xc2.grad) < 2e-2
+
+
+def test_loss():
+    from eventgrad_amd.ops import functional as O
+    torch.manual_seed(5)
+    logits = torch.randn(64, 10)
+    tgt = torch.randint(0, 10, (64,))
+    lg = bq(logits).to(DEV).to(torch.bfloat16).requires_grad_(True)
+    tg = tgt.to(DEV)
+    loss_g = O.nll_of_logits(lg, tg)
+    lc = bq(logits).requires_grad_(True)
+    loss_c = F.nll_loss(F.log_softmax(lc, 1), tgt)
+    assert abs(loss_g.item() - loss_c.item()) < 2e-2
+    loss_g.backward()
+    loss_c.backward()
+    assert rel_err(lg.grad.float().cpu(), lc.grad) < 2e-2
+
+
+def test_dropout_stats():
+    from eventgrad_amd.ops.backend import native
+    x = torch.ones(100000, device=DEV).to(torch.bfloat16)
+    y, mask = native().dropout_fwd(x, 0.5, 1234, False)
+    keep = mask.float().mean().item()
+    assert 0.47 < keep < 0.53
+    # kept elements scaled by 1/(1-p)
+    assert abs(y.float().sum().item() - 2.0 * mask.float().sum().item()) < 10
+    # deterministic for same seed
+    y2, mask2 = native().dropout_fwd(x, 0.5, 1234, False)
+    assert torch.equal(mask, mask2)
+
+
+# ------------------------------------------------------- engine kernels ----
+
+def _mk_flat(sz=5, seed=0):
+    torch.manual_seed(seed)
+    numels = [100, 64, 1280, 7, 333][:sz]
+    starts, off = [], 0
+    for n in numels:
+        starts.append(off)
+        off += (n + 63) // 64 * 64
+    total = off
+    flat = torch.zeros(total, device=DEV)
+    for s, n in zip(starts, numels):
+        flat[s:s + n] = torch.randn(n, device=DEV)
+    st = torch.tensor(starts, dtype=torch.int64, device=DEV)
+    nu = torch.tensor(numels, dtype=torch.int64, device=DEV)
+    return flat, st, nu, starts, numels
+
+
+def test_seg_sqnorms():
+    flat, st, nu, starts, numels = _mk_flat()
+    out = core().seg_sqnorms(flat, st, nu)
+    for i, (s, n) in enumerate(zip(starts, numels)):
+        ref = flat[s:s + n].square().sum().item()
+        assert abs(out[i].item() - ref) / (ref + 1e-9) < 1e-5
+
+
+def test_sgd_step_norm_vs_cpu():
+    flat, st, nu, starts, numels = _mk_flat(seed=7)
+    grad = torch.randn_like(flat)
+    # zero the pad gaps of grad like the real flat-grad buffer
+    mask = torch.zeros_like(flat)
+    for s, n in zip(starts, numels):
+        mask[s:s + n] = 1
+    grad *= mask
+    mom = torch.randn_like(flat) * mask
+    p_ref = flat.cpu().clone()
+    g_ref = grad.cpu().clone()
+    m_ref = mom.cpu().clone()
+    norms = core().sgd_step_norm(flat, grad, mom, st, nu, 0.1, 0.9, 0.0)
+    m_ref.mul_(0.9).add_(g_ref)
+    p_ref.add_(m_ref, alpha=-0.1)
+    assert rel_err(flat.cpu(), p_ref) < 1e-6
+    assert rel_err(mom.cpu(), m_ref) < 1e-6
+    for i, (s, n) in enumerate(zip(starts, numels)):
+        ref = p_ref[s:s + n].square().sum().item()
+        assert abs(norms[i].item() - ref) / (ref + 1e-9) < 1e-5
+
+
+def test_avg3():
+    flat, *_ = _mk_flat(seed=8)
+    l = torch.randn_like(flat)
+    r = torch.randn_like(flat)
+    ref = (flat + l + r) / 3
+    core().avg3(flat, l, r)
+    assert rel_err(flat, ref) < 1e-6
+
+
+def test_gather_scatter_segments():
+    flat, st, nu, starts, numels = _mk_flat(seed=9)
+    fired = [0, 2, 4]
+    src = torch.tensor([starts[i] for i in fired], device=DEV)
+    sizes = [numels[i] for i in fired]
+    cum = np.concatenate([[0], np.cumsum(sizes)])
+    offs = torch.tensor(cum, dtype=torch.int64, device=DEV)
+    payload = core().gather_segments(flat, src, offs, int(cum[-1]))
+    ref = torch.cat([flat[starts[i]:starts[i] + numels[i]] for i in fired])
+    assert torch.equal(payload, ref)
+    inbox = torch.zeros_like(flat)
+    core().scatter_segments(payload, src, offs, inbox)
+    for i in fired:
+        assert torch.equal(inbox[starts[i]:starts[i] + numels[i]],
+                           flat[starts[i]:starts[i] + numels[i]])
+
+
+def test_trigger_update_matches_host_controller():
+    from eventgrad_amd.parallel.controller import TriggerController
+    sz, H = 8, 2
+    host = TriggerController(sz, adaptive=True, horizon=1.05, constant=5e-4,
+                             sent_history=H, initial_comm_passes=3)
+    thres = torch.zeros(sz, device=DEV)
+    lsn = torch.zeros(sz, device=DEV)
+    lsi = torch.zeros(sz, device=DEV)
+    slopes = torch.zeros(sz * H, device=DEV)
+    nev = torch.zeros(1, dtype=torch.int32, device=DEV)
+    rng = np.random.default_rng(0)
+    for p in range(1, 30):
+        norms = rng.random(sz).astype(np.float32) * (1 + p / 10)
+        mh = host.step(norms, p)
+        nd = torch.tensor(norms**2, device=DEV)
+        mg = core().trigger_update(nd, thres, lsn, lsi, slopes, nev, p, True,
+                                   1.05, 5e-4, 3, False)
+        assert (mg.cpu().numpy().astype(bool) == mh).all(), p
+        np.testing.assert_allclose(thres.cpu().numpy(), host.thres,
+                                   rtol=1e-4, atol=1e-7)
+    assert int(nev.item()) == host.num_events
+
+
+def test_topk_absdiff():
+    torch.manual_seed(11)
+    n, k = 10000, 77
+    x = torch.randn(n, device=DEV)
+    prev = torch.randn(n, device=DEV)
+    prev_copy = prev.clone()
+    vals, idx = core().topk_absdiff(x, prev, k)
+    diff = (x - prev_copy).abs()
+    ref_vals, ref_idx = torch.topk(diff, k)
+    # same selected |diff| multiset (ties may reorder)
+    got = diff[idx.long()].sort().values
+    assert torch.allclose(got, ref_vals.sort().values, atol=1e-6)
+    # vals are x at the selected indices; prev updated there
+    assert torch.allclose(vals, x[idx.long()])
+    assert torch.allclose(prev[idx.long()], x[idx.long()])
+    untouched = torch.ones(n, dtype=torch.bool, device=DEV)
+    untouched[idx.long()] = False
+    assert torch.equal(prev[untouched], prev_copy[untouched])

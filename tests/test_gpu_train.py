@@ -1,0 +1,106 @@
+"""GPU end-to-end: model fwd/bwd parity vs CPU fp32, serial training smoke,
+and the native-extension-actually-loaded guard."""
+
+import numpy as np
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+
+def test_native_extension_loaded_from_tree():
+    import eventgrad_amd._core as c
+    assert "eventgrad_amd" in c.__file__
+    # ops on GPU must route through it: a conv on cuda with the module
+    # present must not raise
+    from eventgrad_amd.ops import functional as O
+    x = torch.randn(1, 4, 4, 8, device="cuda").to(torch.bfloat16)
+    w = torch.randn(8, 8, 3, 3, device="cuda")
+    y = O.conv2d(x, w, None, 1, 1)
+    assert y.shape == (1, 4, 4, 8)
+
+
+@pytest.mark.parametrize("model_name", ["cnn2", "resnet18q"])
+def test_model_fwd_bwd_parity(model_name):
+    """GPU bf16 model vs CPU fp32 model with identical weights."""
+    from eventgrad_amd.models import build_model
+    from eventgrad_amd.ops import functional as O
+
+    torch.manual_seed(0)
+    m_cpu = build_model(model_name)
+    torch.manual_seed(0)
+    m_gpu = build_model(model_name).to("cuda")
+    # identical weights
+    for pc, pg in zip(m_cpu.parameters(), m_gpu.parameters()):
+        assert torch.allclose(pc, pg.cpu())
+
+    shape = (8, 1, 28, 28) if model_name == "cnn2" else (8, 3, 32, 32)
+    torch.manual_seed(1)
+    x = torch.randn(shape)
+    y = torch.randint(0, 10, (shape[0],))
+
+    m_cpu.train()
+    m_gpu.train()
+    # disable dropout randomness differences by eval'ing dropout via p=0:
+    # CNN2 uses dropout; parity check therefore runs in eval-forward but
+    # with training-mode BN? Simplest: compare in train mode for resnet
+    # (no dropout) and eval mode for cnn2.
+    if model_name == "cnn2":
+        m_cpu.eval()
+        m_gpu.eval()
+        lc = m_cpu(x)
+        lg = m_gpu(x.cuda())
+        err = (lg.float().cpu() - lc).norm() / lc.norm()
+        assert err < 0.05, err.item()
+        return
+
+    logits_c = m_cpu(x)
+    loss_c = O.nll_of_logits(logits_c, y)
+    loss_c.backward()
+    logits_g = m_gpu(x.cuda())
+    loss_g = O.nll_of_logits(logits_g, y.cuda())
+    loss_g.backward()
+    assert abs(loss_g.item() - loss_c.item()) / abs(loss_c.item()) < 0.05
+
+    # gradient direction must agree (bf16 noise accumulates over 26 convs)
+    cos_all = []
+    for (n, pc), pg in zip(m_cpu.named_parameters(), m_gpu.parameters()):
+        gc = pc.grad.flatten()
+        gg = pg.grad.cpu().flatten()
+        cos = torch.dot(gc, gg) / (gc.norm() * gg.norm() + 1e-12)
+        cos_all.append(cos.item())
+    assert np.mean(cos_all) > 0.97, np.mean(cos_all)
+    assert min(cos_all) > 0.85, (min(cos_all),
+                                 [n for (n, _), c in
+                                  zip(m_cpu.named_parameters(), cos_all)
+                                  if c < 0.9])
+
+
+def test_serial_training_learns_gpu():
+    from eventgrad_amd.config import DataConfig, OptimConfig, RunConfig
+    from eventgrad_amd.train.trainer import Trainer
+
+    cfg = RunConfig(
+        mode="serial", model="cnn2", epochs=3, device="cuda",
+        data=DataConfig(dataset="synthetic-mnist", batch_size=64,
+                        synthetic_train_samples=512,
+                        synthetic_test_samples=128, synthetic_noise=0.3),
+        optim=OptimConfig(lr=0.05), eval_at_end=False)
+    tr = Trainer(cfg)
+    m = tr.train()
+    assert np.isfinite(m.final_train_loss)
+    assert m.epoch_train_acc[-1] > m.epoch_train_acc[0]
+
+
+def test_resnet_training_step_gpu():
+    from eventgrad_amd.config import DataConfig, OptimConfig, RunConfig
+    from eventgrad_amd.train.trainer import Trainer
+
+    cfg = RunConfig(
+        mode="serial", model="resnet18q", epochs=1, device="cuda",
+        data=DataConfig(dataset="synthetic", batch_size=32,
+                        synthetic_train_samples=64, synthetic_test_samples=32),
+        optim=OptimConfig(lr=0.01, momentum=0.9), eval_at_end=False)
+    tr = Trainer(cfg)
+    m = tr.train()
+    assert np.isfinite(m.final_train_loss)
