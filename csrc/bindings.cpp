@@ -24,6 +24,9 @@ torch::Tensor add_relu_fwd(torch::Tensor, torch::Tensor);
 std::vector<torch::Tensor> dropout_fwd(torch::Tensor, double, long, bool);
 torch::Tensor dropout_bwd(torch::Tensor, torch::Tensor, double, bool);
 torch::Tensor channel_sum(torch::Tensor);
+torch::Tensor oihw_to_krsc(torch::Tensor);
+torch::Tensor krsc_to_crsk(torch::Tensor);
+torch::Tensor krsc_to_oihw(torch::Tensor);
 // gemm.hip
 torch::Tensor gemm_bias(torch::Tensor, torch::Tensor, torch::Tensor, bool);
 // conv.hip
@@ -66,6 +69,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("dropout_fwd", &eg::dropout_fwd);
   m.def("dropout_bwd", &eg::dropout_bwd);
   m.def("channel_sum", &eg::channel_sum);
+  m.def("oihw_to_krsc", &eg::oihw_to_krsc);
+  m.def("krsc_to_crsk", &eg::krsc_to_crsk);
+  m.def("krsc_to_oihw", &eg::krsc_to_oihw);
   m.def("gemm_bias", &eg::gemm_bias);
   m.def("conv2d_fwd", &eg::conv2d_fwd);
   m.def("conv2d_dgrad", &eg::conv2d_dgrad);

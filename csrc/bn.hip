@@ -10,7 +10,48 @@ namespace eg {
 
 namespace bn {
 
-// per-channel sum & sumsq of x (bf16 [rows, C]) -> fp32 [C] each
+// per-channel sum & sumsq of x (bf16 [rows, C]) -> fp32 [C] each.
+// Vectorized form (C % 8 == 0, C <= 1024): each lane owns 8 consecutive
+// channels via 16-B loads; a 256-thread block covers 256*8/C rows per step
+// (guide §6 G13 — scalar bf16 channel reads ran at ~6% of HBM peak).
+constexpr int BN_MAXC = 1024;
+
+__global__ void stats_kernel_v(const bf16* __restrict__ x,
+                               float* __restrict__ s, float* __restrict__ sq,
+                               long rows, int c) {
+  __shared__ float ls[BN_MAXC], lsq[BN_MAXC];
+  for (int i = threadIdx.x; i < c; i += blockDim.x) {
+    ls[i] = 0.f;
+    lsq[i] = 0.f;
+  }
+  __syncthreads();
+  const int lpr = c >> 3;                 // lanes per row
+  const int c0 = (threadIdx.x % lpr) * 8;
+  const int rpb = blockDim.x / lpr;       // rows per block-step
+  float a[8] = {}, b[8] = {};
+  for (long r = (long)blockIdx.x * rpb + threadIdx.x / lpr; r < rows;
+       r += (long)gridDim.x * rpb) {
+    s16x8 v = *reinterpret_cast<const s16x8*>(&x[r * c + c0]);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float f = b2f(__ushort_as_bfloat16((unsigned short)v[j]));
+      a[j] += f;
+      b[j] += f * f;
+    }
+  }
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    atomicAdd(&ls[c0 + j], a[j]);
+    atomicAdd(&lsq[c0 + j], b[j]);
+  }
+  __syncthreads();
+  for (int i = threadIdx.x; i < c; i += blockDim.x) {
+    atomicAdd(&s[i], ls[i]);
+    atomicAdd(&sq[i], lsq[i]);
+  }
+}
+
+// scalar fallback for C % 8 != 0
 __global__ void stats_kernel(const bf16* __restrict__ x, float* __restrict__ s,
                              float* __restrict__ sq, long rows, int c) {
   int ch = blockIdx.x * 64 + (threadIdx.x & 63);
@@ -74,7 +115,63 @@ __global__ void norm_kernel(const bf16* __restrict__ x, bf16* __restrict__ y,
   }
 }
 
-// backward reductions: sum(dy'), sum(dy' * xhat) per channel
+// vectorized backward reductions (C % 8 == 0): sum(dy'), sum(dy'*xhat)
+__global__ void bwd_stats_kernel_v(const bf16* __restrict__ dy,
+                                   const bf16* __restrict__ x,
+                                   const bf16* __restrict__ y,
+                                   const float* __restrict__ mean,
+                                   const float* __restrict__ invstd,
+                                   float* __restrict__ sum_dy,
+                                   float* __restrict__ sum_dyx, long rows,
+                                   int c, int relu) {
+  __shared__ float ls[BN_MAXC], lsq[BN_MAXC];
+  for (int i = threadIdx.x; i < c; i += blockDim.x) {
+    ls[i] = 0.f;
+    lsq[i] = 0.f;
+  }
+  __syncthreads();
+  const int lpr = c >> 3;
+  const int c0 = (threadIdx.x % lpr) * 8;
+  const int rpb = blockDim.x / lpr;
+  float m[8], is[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    m[j] = mean[c0 + j];
+    is[j] = invstd[c0 + j];
+  }
+  float a[8] = {}, b[8] = {};
+  for (long r = (long)blockIdx.x * rpb + threadIdx.x / lpr; r < rows;
+       r += (long)gridDim.x * rpb) {
+    long base = r * c + c0;
+    s16x8 vg = *reinterpret_cast<const s16x8*>(&dy[base]);
+    s16x8 vx = *reinterpret_cast<const s16x8*>(&x[base]);
+    s16x8 vy;
+    if (relu) vy = *reinterpret_cast<const s16x8*>(&y[base]);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float g = b2f(__ushort_as_bfloat16((unsigned short)vg[j]));
+      if (relu &&
+          b2f(__ushort_as_bfloat16((unsigned short)vy[j])) <= 0.f)
+        g = 0.f;
+      float xh =
+          (b2f(__ushort_as_bfloat16((unsigned short)vx[j])) - m[j]) * is[j];
+      a[j] += g;
+      b[j] += g * xh;
+    }
+  }
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    atomicAdd(&ls[c0 + j], a[j]);
+    atomicAdd(&lsq[c0 + j], b[j]);
+  }
+  __syncthreads();
+  for (int i = threadIdx.x; i < c; i += blockDim.x) {
+    atomicAdd(&sum_dy[i], ls[i]);
+    atomicAdd(&sum_dyx[i], lsq[i]);
+  }
+}
+
+// backward reductions: sum(dy'), sum(dy' * xhat) per channel (fallback)
 __global__ void bwd_stats_kernel(const bf16* __restrict__ dy,
                                  const bf16* __restrict__ x,
                                  const bf16* __restrict__ y,
@@ -147,10 +244,19 @@ std::vector<torch::Tensor> bn_fwd(torch::Tensor x, torch::Tensor gamma,
   if (training) {
     auto s = torch::zeros({c}, f32);
     auto sq = torch::zeros({c}, f32);
-    dim3 grid((c + 63) / 64, (unsigned)std::min<long>((rows + 3) / 4, 256L));
-    bn::stats_kernel<<<grid, 256, 0, stream>>>(
-        (const bf16*)x.data_ptr(), s.data_ptr<float>(), sq.data_ptr<float>(),
-        rows, c);
+    if (c % 8 == 0 && c <= bn::BN_MAXC) {
+      int rpb = 256 / (c / 8);
+      int grid = (int)std::min<long>((rows + rpb - 1) / rpb, 1024L);
+      bn::stats_kernel_v<<<grid, 256, 0, stream>>>(
+          (const bf16*)x.data_ptr(), s.data_ptr<float>(),
+          sq.data_ptr<float>(), rows, c);
+    } else {
+      dim3 grid((c + 63) / 64,
+                (unsigned)std::min<long>((rows + 3) / 4, 256L));
+      bn::stats_kernel<<<grid, 256, 0, stream>>>(
+          (const bf16*)x.data_ptr(), s.data_ptr<float>(),
+          sq.data_ptr<float>(), rows, c);
+    }
     bn::finalize_kernel<<<ceil_div(c, 128), 128, 0, stream>>>(
         s.data_ptr<float>(), sq.data_ptr<float>(), mean.data_ptr<float>(),
         invstd.data_ptr<float>(), running_mean.data_ptr<float>(),
@@ -183,12 +289,22 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor x,
   auto sum_dy = torch::zeros({c}, f32);
   auto sum_dyx = torch::zeros({c}, f32);
   auto stream = cur_stream();
-  dim3 grid((c + 63) / 64, (unsigned)std::min<long>((rows + 3) / 4, 256L));
-  bn::bwd_stats_kernel<<<grid, 256, 0, stream>>>(
-      (const bf16*)dy.data_ptr(), (const bf16*)x.data_ptr(),
-      (const bf16*)y.data_ptr(), mean.data_ptr<float>(),
-      invstd.data_ptr<float>(), sum_dy.data_ptr<float>(),
-      sum_dyx.data_ptr<float>(), rows, c, relu ? 1 : 0);
+  if (c % 8 == 0 && c <= bn::BN_MAXC) {
+    int rpb = 256 / (c / 8);
+    int grid = (int)std::min<long>((rows + rpb - 1) / rpb, 1024L);
+    bn::bwd_stats_kernel_v<<<grid, 256, 0, stream>>>(
+        (const bf16*)dy.data_ptr(), (const bf16*)x.data_ptr(),
+        (const bf16*)y.data_ptr(), mean.data_ptr<float>(),
+        invstd.data_ptr<float>(), sum_dy.data_ptr<float>(),
+        sum_dyx.data_ptr<float>(), rows, c, relu ? 1 : 0);
+  } else {
+    dim3 grid((c + 63) / 64, (unsigned)std::min<long>((rows + 3) / 4, 256L));
+    bn::bwd_stats_kernel<<<grid, 256, 0, stream>>>(
+        (const bf16*)dy.data_ptr(), (const bf16*)x.data_ptr(),
+        (const bf16*)y.data_ptr(), mean.data_ptr<float>(),
+        invstd.data_ptr<float>(), sum_dy.data_ptr<float>(),
+        sum_dyx.data_ptr<float>(), rows, c, relu ? 1 : 0);
+  }
   auto dx = torch::empty_like(dy);
   int grid1 = (int)std::min<long>((n + 255) / 256, 4096L);
   bn::bwd_dx_kernel<<<grid1, 256, 0, stream>>>(

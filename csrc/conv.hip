@@ -159,51 +159,65 @@ __device__ __forceinline__ void row8(const bf16* g, long rows, long cols,
 }
 
 // ---- fwd & dgrad share the main loop (template on the A-stager) ----------
+// 128x64 block tile, BK=64, 8 waves (4x2) of 32x32, 512 threads.
+
+constexpr int CBM = 128, CBN = 64, CBK = 64;
+constexpr int CLDK = CBK + 8;
 
 template <int MODE>  // 0 = fwd, 1 = dgrad
-__global__ __launch_bounds__(256) void conv_mm_kernel(
+__global__ __launch_bounds__(512) void conv_mm_kernel(
     const bf16* __restrict__ Asrc, const bf16* __restrict__ B,
     const float* __restrict__ bias, bf16* __restrict__ out, Geom g,
     long M, long N, long RED, int has_bias) {
-  __shared__ __bf16 sA[BM * LDK];
-  __shared__ __bf16 sB[BN * LDK];
+  __shared__ __bf16 sA[CBM * CLDK];
+  __shared__ __bf16 sB[CBN * CLDK];
 
-  const long m0 = (long)blockIdx.x * BM;
-  const long n0 = (long)blockIdx.y * BN;
+  const long m0 = (long)blockIdx.x * CBM;
+  const long n0 = (long)blockIdx.y * CBN;
   const int t = threadIdx.x;
   const int lane = t & 63;
-  const int wave = t >> 6;
+  const int wave = t >> 6;          // 8 waves: wr in 0..3, wc in 0..1
   const int wr = wave >> 1, wc = wave & 1;
-  const int lr = t >> 2;
-  const int lk = (t & 3) * 8;
+  // A staging: 512 thr x 2 vec8 = 128x64; B staging: 512 thr x 1 vec8 = 64x64
+  const int lra = t >> 2;            // 0..127
+  const int lka = (t & 3) * 8;       // 0..24
+  const int lrb = t >> 3;            // 0..63
+  const int lkb = (t & 7) * 8;       // 0..56
 
   f32x4 acc[2][2] = {};
 
-  for (long k0 = 0; k0 < RED; k0 += BK) {
-    __bf16 ra[8], rb[8];
+  for (long k0 = 0; k0 < RED; k0 += CBK) {
+    __bf16 ra0[8], ra1[8], rb[8];
     if (MODE == 0) {
-      stage_im2col8(Asrc, g, m0 + lr, (int)(k0 + lk), ra);
+      stage_im2col8(Asrc, g, m0 + lra, (int)(k0 + lka), ra0);
+      stage_im2col8(Asrc, g, m0 + lra, (int)(k0 + lka + 32), ra1);
     } else {
-      stage_dgrad8(Asrc, g, m0 + lr, (int)(k0 + lk), ra);
+      stage_dgrad8(Asrc, g, m0 + lra, (int)(k0 + lka), ra0);
+      stage_dgrad8(Asrc, g, m0 + lra, (int)(k0 + lka + 32), ra1);
     }
-    row8(B, N, RED, n0 + lr, k0 + lk, rb);
+    row8(B, N, RED, n0 + lrb, k0 + lkb, rb);
     __syncthreads();
-    *reinterpret_cast<bf16x8*>(&sA[lr * LDK + lk]) =
-        *reinterpret_cast<bf16x8*>(ra);
-    *reinterpret_cast<bf16x8*>(&sB[lr * LDK + lk]) =
+    *reinterpret_cast<bf16x8*>(&sA[lra * CLDK + lka]) =
+        *reinterpret_cast<bf16x8*>(ra0);
+    *reinterpret_cast<bf16x8*>(&sA[lra * CLDK + lka + 32]) =
+        *reinterpret_cast<bf16x8*>(ra1);
+    *reinterpret_cast<bf16x8*>(&sB[lrb * CLDK + lkb]) =
         *reinterpret_cast<bf16x8*>(rb);
     __syncthreads();
 
-    const int kf = (lane >> 4) * 8;
     const int ml = lane & 15;
 #pragma unroll
-    for (int fi = 0; fi < 2; ++fi) {
-      bf16x8 af = lds8(&sA[(wr * 32 + fi * 16 + ml) * LDK + kf]);
+    for (int ks = 0; ks < 2; ++ks) {
+      const int kf = ks * 32 + (lane >> 4) * 8;
 #pragma unroll
-      for (int fj = 0; fj < 2; ++fj) {
-        bf16x8 bfr = lds8(&sB[(wc * 32 + fj * 16 + ml) * LDK + kf]);
-        acc[fi][fj] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            af, bfr, acc[fi][fj], 0, 0, 0);
+      for (int fi = 0; fi < 2; ++fi) {
+        bf16x8 af = lds8(&sA[(wr * 32 + fi * 16 + ml) * CLDK + kf]);
+#pragma unroll
+        for (int fj = 0; fj < 2; ++fj) {
+          bf16x8 bfr = lds8(&sB[(wc * 32 + fj * 16 + ml) * CLDK + kf]);
+          acc[fi][fj] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af, bfr, acc[fi][fj], 0, 0, 0);
+        }
       }
     }
   }
@@ -336,8 +350,8 @@ torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor bias,
   long M = (long)g.N * g.Ho * g.Wo;
   long RED = (long)g.R * g.S * g.C;
   bool has_bias = bias.numel() > 0;
-  dim3 grid(ceil_div(M, conv::BM), ceil_div(g.K, conv::BN));
-  conv::conv_mm_kernel<0><<<grid, 256, 0, cur_stream()>>>(
+  dim3 grid(ceil_div(M, conv::CBM), ceil_div(g.K, conv::CBN));
+  conv::conv_mm_kernel<0><<<grid, 512, 0, cur_stream()>>>(
       (const bf16*)x.data_ptr(), (const bf16*)w.data_ptr(),
       has_bias ? bias.data_ptr<float>() : nullptr, (bf16*)y.data_ptr(), g,
       M, g.K, RED, has_bias ? 1 : 0);
@@ -358,8 +372,8 @@ torch::Tensor conv2d_dgrad(torch::Tensor dy, torch::Tensor wt, long stride,
   auto dx = torch::empty({g.N, g.H, g.W, g.C}, dy.options());
   long M = (long)g.N * g.H * g.W;
   long RED = (long)R * S * K;
-  dim3 grid(ceil_div(M, conv::BM), ceil_div(C, conv::BN));
-  conv::conv_mm_kernel<1><<<grid, 256, 0, cur_stream()>>>(
+  dim3 grid(ceil_div(M, conv::CBM), ceil_div(C, conv::CBN));
+  conv::conv_mm_kernel<1><<<grid, 512, 0, cur_stream()>>>(
       (const bf16*)dy.data_ptr(), (const bf16*)wt.data_ptr(), nullptr,
       (bf16*)dx.data_ptr(), g, M, C, RED, 0);
   return dx;
@@ -376,9 +390,10 @@ torch::Tensor conv2d_wgrad(torch::Tensor x, torch::Tensor dy, long R, long S,
   long RED = (long)R * S * g.C;
   auto dw = torch::zeros({(long)g.K, RED}, x.options().dtype(torch::kFloat32));
   // split the NP reduction across blocks for parallelism; fp32 atomics
-  long target_blocks = 512;
+  long target_blocks = 1024;
   long tiles = (long)ceil_div(g.K, conv::BM) * ceil_div(RED, conv::BN);
-  long zsplit = std::max(1L, std::min(64L, target_blocks / std::max(tiles, 1L)));
+  long zsplit =
+      std::max(1L, std::min(512L, target_blocks / std::max(tiles, 1L)));
   long npslice = (NP + zsplit - 1) / zsplit;
   npslice = ((npslice + conv::BK - 1) / conv::BK) * conv::BK;
   zsplit = (NP + npslice - 1) / npslice;

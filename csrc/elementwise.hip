@@ -185,4 +185,102 @@ torch::Tensor channel_sum(torch::Tensor x) {
   return out;
 }
 
+
+// ---- fused conv-weight layout transforms (one kernel per transform) ------
+// OIHW fp32 -> KRSC bf16 (fwd operand), KRSC bf16 -> CRSK bf16 (dgrad
+// operand), KRSC fp32 -> OIHW fp32 (wgrad output). Each replaces a
+// permute+contiguous+cast aten chain (2-3 launches) on the per-step path.
+
+namespace wt {
+
+__global__ void oihw_to_krsc_kernel(const float* __restrict__ w,
+                                    bf16* __restrict__ out, int K, int C,
+                                    int R, int S) {
+  long total = (long)K * C * R * S;
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += stride) {
+    // out index: (((k*R + r)*S + s)*C + c)
+    int c = (int)(i % C);
+    long rem = i / C;
+    int s = (int)(rem % S);
+    rem /= S;
+    int r = (int)(rem % R);
+    int k = (int)(rem / R);
+    out[i] = f2b(w[((long)(k * C + c) * R + r) * S + s]);
+  }
+}
+
+__global__ void krsc_to_crsk_kernel(const bf16* __restrict__ wk,
+                                    bf16* __restrict__ out, int K, int C,
+                                    int R, int S) {
+  long total = (long)K * C * R * S;
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += stride) {
+    // out index: (((c*R + r)*S + s)*K + k)
+    int k = (int)(i % K);
+    long rem = i / K;
+    int s = (int)(rem % S);
+    rem /= S;
+    int r = (int)(rem % R);
+    int c = (int)(rem / R);
+    out[i] = wk[((long)(k * R + r) * S + s) * C + c];
+  }
+}
+
+__global__ void krsc_to_oihw_kernel(const float* __restrict__ dwk,
+                                    float* __restrict__ out, int K, int C,
+                                    int R, int S) {
+  long total = (long)K * C * R * S;
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += stride) {
+    // out (OIHW) index: ((k*C + c)*R + r)*S + s
+    int s = (int)(i % S);
+    long rem = i / S;
+    int r = (int)(rem % R);
+    rem /= R;
+    int c = (int)(rem % C);
+    int k = (int)(rem / C);
+    out[i] = dwk[((long)(k * R + r) * S + s) * C + c];
+  }
+}
+
+}  // namespace wt
+
+static inline int wt_grid(long total) {
+  return (int)std::min<long>((total + 255) / 256, 2048L);
+}
+
+torch::Tensor oihw_to_krsc(torch::Tensor w) {
+  CHECK_IN(w);
+  int K = w.size(0), C = w.size(1), R = w.size(2), S = w.size(3);
+  auto out = torch::empty({K, R, S, C}, w.options().dtype(torch::kBFloat16));
+  long total = (long)K * C * R * S;
+  wt::oihw_to_krsc_kernel<<<wt_grid(total), 256, 0, cur_stream()>>>(
+      w.data_ptr<float>(), (bf16*)out.data_ptr(), K, C, R, S);
+  return out;
+}
+
+torch::Tensor krsc_to_crsk(torch::Tensor wk) {
+  CHECK_IN(wk);
+  int K = wk.size(0), R = wk.size(1), S = wk.size(2), C = wk.size(3);
+  auto out = torch::empty({C, R, S, K}, wk.options());
+  long total = (long)K * C * R * S;
+  wt::krsc_to_crsk_kernel<<<wt_grid(total), 256, 0, cur_stream()>>>(
+      (const bf16*)wk.data_ptr(), (bf16*)out.data_ptr(), K, C, R, S);
+  return out;
+}
+
+torch::Tensor krsc_to_oihw(torch::Tensor dwk) {
+  CHECK_IN(dwk);
+  int K = dwk.size(0), R = dwk.size(1), S = dwk.size(2), C = dwk.size(3);
+  auto out = torch::empty({K, C, R, S}, dwk.options());
+  long total = (long)K * C * R * S;
+  wt::krsc_to_oihw_kernel<<<wt_grid(total), 256, 0, cur_stream()>>>(
+      dwk.data_ptr<float>(), out.data_ptr<float>(), K, C, R, S);
+  return out;
+}
+
 }  // namespace eg

@@ -54,7 +54,7 @@ def flatten_features(x: torch.Tensor) -> torch.Tensor:
 
 def _w_krsc(w: torch.Tensor) -> torch.Tensor:
     """OIHW fp32 -> KRSC bf16 ([K,R,S,C]) for the fwd implicit GEMM."""
-    return w.permute(0, 2, 3, 1).contiguous().to(torch.bfloat16)
+    return native().oihw_to_krsc(w.detach().contiguous())
 
 
 # --------------------------------------------------------------------------
@@ -82,13 +82,13 @@ class _ConvNHWC(torch.autograd.Function):
         dx = dw = db = None
         if ctx.needs_input_grad[0]:
             # dgrad weight layout: [C,R,S,K] from [K,R,S,C]
-            w_crsk = wk.permute(3, 1, 2, 0).contiguous()
+            w_crsk = core.krsc_to_crsk(wk)
             dx = core.conv2d_dgrad(dy, w_crsk, ctx.stride, ctx.padding,
                                    ctx.hw[0], ctx.hw[1])
         if ctx.needs_input_grad[1]:
             dw_krsc = core.conv2d_wgrad(x, dy, wk.shape[1], wk.shape[2],
                                         ctx.stride, ctx.padding)
-            dw = dw_krsc.permute(0, 3, 1, 2).contiguous()  # KRSC -> OIHW fp32
+            dw = core.krsc_to_oihw(dw_krsc)  # KRSC fp32 -> OIHW fp32
         if ctx.has_bias and ctx.needs_input_grad[2]:
             db = core.channel_sum(dy)
         return dx, dw, db, None, None
