@@ -242,8 +242,9 @@ std::vector<torch::Tensor> bn_fwd(torch::Tensor x, torch::Tensor gamma,
   auto invstd = torch::empty({c}, f32);
   auto stream = cur_stream();
   if (training) {
-    auto s = torch::zeros({c}, f32);
-    auto sq = torch::zeros({c}, f32);
+    auto ssq = torch::zeros({2 * c}, f32);
+    auto s = ssq.narrow(0, 0, c);
+    auto sq = ssq.narrow(0, c, c);
     if (c % 8 == 0 && c <= bn::BN_MAXC) {
       int rpb = 256 / (c / 8);
       int grid = (int)std::min<long>((rows + rpb - 1) / rpb, 1024L);
@@ -286,8 +287,9 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor x,
   long rows = x.numel() / c;
   long n = x.numel();
   auto f32 = x.options().dtype(torch::kFloat32);
-  auto sum_dy = torch::zeros({c}, f32);
-  auto sum_dyx = torch::zeros({c}, f32);
+  auto sums = torch::zeros({2 * c}, f32);
+  auto sum_dy = sums.narrow(0, 0, c);
+  auto sum_dyx = sums.narrow(0, c, c);
   auto stream = cur_stream();
   if (c % 8 == 0 && c <= bn::BN_MAXC) {
     int rpb = 256 / (c / 8);

@@ -9,12 +9,16 @@
 // GEMM views (all row-major, K-innermost "NT" tiles like gemm.hip):
 //  fwd   y[np, k]  = im2col[np, rsc]  @ w[k, rsc]^T          np=(n,ho,wo)
 //  dgrad dx[np, c] = col(dy)[np, rsk] @ wT[c, rsk]^T         np=(n,h,w)
-//        (full-corr with rotated kernel: dy upsampled by stride)
 //  wgrad dw[k, rsc] = dy[np, k]^T @ im2col[np, rsc]          (TN, split-NP
 //        across blocks, fp32 atomic accumulate)
 //
-// Tiles: 64x64xBK32, 4 waves (2x2) of 2x2 v_mfma_f32_16x16x32_bf16
-// fragments; LDS rows padded +8 bf16 against b128 bank conflicts.
+// Address-math discipline (the first profile showed these kernels
+// VALU-bound on div/mod chains): each thread's output-pixel decomposition
+// is hoisted OUT of the k-loop, and the reduction index (r,s,c)/(r,s,k)
+// advances by an incremental cursor (add + wrap) instead of div/mod —
+// FAST template path, used whenever the innermost dim is a multiple of 8
+// (every ResNet conv). The generic slow path remains for C=3 stems and
+// the small CNN channel counts.
 
 #include "common.h"
 
@@ -22,8 +26,8 @@ namespace eg {
 
 namespace conv {
 
-constexpr int BM = 64, BN = 64, BK = 32;
-constexpr int LDK = BK + 8;
+constexpr int CBM = 128, CBN = 64, CBK = 64;
+constexpr int CLDK = CBK + 8;
 
 struct Geom {
   int N, H, W, C;     // input
@@ -36,111 +40,56 @@ __device__ __forceinline__ bf16x8 lds8(const __bf16* p) {
   return *reinterpret_cast<const bf16x8*>(p);
 }
 
-// stage one vec8 of the im2col A-operand: row m (output pixel), red r0..r0+7.
-__device__ __forceinline__ void stage_im2col8(const bf16* __restrict__ x,
-                                              const Geom g, long m, int red,
-                                              __bf16* dst) {
-  long NP = (long)g.N * g.Ho * g.Wo;
-  if (m >= NP) {
+__device__ __forceinline__ void zero8(__bf16* dst) {
 #pragma unroll
-    for (int j = 0; j < 8; ++j) dst[j] = (__bf16)0.f;
-    return;
-  }
-  int n = (int)(m / ((long)g.Ho * g.Wo));
-  int rem = (int)(m % ((long)g.Ho * g.Wo));
-  int ho = rem / g.Wo, wo = rem % g.Wo;
-  int rsc = red;
-  int c0 = rsc % g.C;
-  int rs = rsc / g.C;
-  int r = rs / g.S, s = rs % g.S;
-  int h = ho * g.stride - g.pad + r;
-  int w = wo * g.stride - g.pad + s;
+  for (int j = 0; j < 8; ++j) dst[j] = (__bf16)0.f;
+}
+
+// ---- generic (slow) stagers: per-element guarded loads -------------------
+
+__device__ __forceinline__ void stage_im2col_slow(const bf16* __restrict__ x,
+                                                  const Geom& g, bool mvalid,
+                                                  int n, int ho, int wo,
+                                                  int red, __bf16* dst) {
   int RSC = g.R * g.S * g.C;
-  bool fast = (c0 + 8 <= g.C) && (red + 8 <= RSC);
-  if (fast) {
-    bool valid = (unsigned)h < (unsigned)g.H && (unsigned)w < (unsigned)g.W;
-    long off = (((long)n * g.H + h) * g.W + w) * g.C + c0;
-    if (valid && (off & 7) == 0) {
-      *reinterpret_cast<s16x8*>(dst) = *reinterpret_cast<const s16x8*>(x + off);
-    } else if (valid) {
 #pragma unroll
-      for (int j = 0; j < 8; ++j) dst[j] = (__bf16)b2f(x[off + j]);
-    } else {
-#pragma unroll
-      for (int j = 0; j < 8; ++j) dst[j] = (__bf16)0.f;
+  for (int j = 0; j < 8; ++j) {
+    int rr = red + j;
+    float v = 0.f;
+    if (mvalid && rr < RSC) {
+      int c = rr % g.C;
+      int rs2 = rr / g.C;
+      int r2 = rs2 / g.S, s2 = rs2 % g.S;
+      int h2 = ho * g.stride - g.pad + r2;
+      int w2 = wo * g.stride - g.pad + s2;
+      if ((unsigned)h2 < (unsigned)g.H && (unsigned)w2 < (unsigned)g.W)
+        v = b2f(x[(((long)n * g.H + h2) * g.W + w2) * g.C + c]);
     }
-  } else {
-#pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      int rr = red + j;
-      float v = 0.f;
-      if (rr < RSC) {
-        int c = rr % g.C;
-        int rs2 = rr / g.C;
-        int r2 = rs2 / g.S, s2 = rs2 % g.S;
-        int h2 = ho * g.stride - g.pad + r2;
-        int w2 = wo * g.stride - g.pad + s2;
-        if ((unsigned)h2 < (unsigned)g.H && (unsigned)w2 < (unsigned)g.W)
-          v = b2f(x[(((long)n * g.H + h2) * g.W + w2) * g.C + c]);
-      }
-      dst[j] = (__bf16)v;
-    }
+    dst[j] = (__bf16)v;
   }
 }
 
-// stage one vec8 of the dgrad A-operand: row m = input pixel (n,h,w),
-// red = (r, s, kk) over the OUTPUT-grad channels.
-__device__ __forceinline__ void stage_dgrad8(const bf16* __restrict__ dy,
-                                             const Geom g, long m, int red,
-                                             __bf16* dst) {
-  long NP = (long)g.N * g.H * g.W;  // over INPUT pixels
+__device__ __forceinline__ void stage_dgrad_slow(const bf16* __restrict__ dy,
+                                                 const Geom& g, bool mvalid,
+                                                 int n, int h, int w, int red,
+                                                 __bf16* dst) {
   int RSK = g.R * g.S * g.K;
-  if (m >= NP) {
 #pragma unroll
-    for (int j = 0; j < 8; ++j) dst[j] = (__bf16)0.f;
-    return;
-  }
-  int n = (int)(m / ((long)g.H * g.W));
-  int rem = (int)(m % ((long)g.H * g.W));
-  int h = rem / g.W, w = rem % g.W;
-  int k0 = red % g.K;
-  bool fast = (k0 + 8 <= g.K) && (red + 8 <= RSK);
-  if (fast) {
-    int rs = red / g.K;
-    int r = rs / g.S, s = rs % g.S;
-    int hq = h + g.pad - r, wq = w + g.pad - s;
-    bool valid = hq >= 0 && wq >= 0 && hq % g.stride == 0 &&
-                 wq % g.stride == 0;
-    int ho = hq / g.stride, wo = wq / g.stride;
-    valid = valid && ho < g.Ho && wo < g.Wo;
-    long off = (((long)n * g.Ho + ho) * g.Wo + wo) * g.K + k0;
-    if (valid && (off & 7) == 0) {
-      *reinterpret_cast<s16x8*>(dst) = *reinterpret_cast<const s16x8*>(dy + off);
-    } else if (valid) {
-#pragma unroll
-      for (int j = 0; j < 8; ++j) dst[j] = (__bf16)b2f(dy[off + j]);
-    } else {
-#pragma unroll
-      for (int j = 0; j < 8; ++j) dst[j] = (__bf16)0.f;
-    }
-  } else {
-#pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      int rr = red + j;
-      float v = 0.f;
-      if (rr < RSK) {
-        int kk = rr % g.K;
-        int rs = rr / g.K;
-        int r = rs / g.S, s = rs % g.S;
-        int hq = h + g.pad - r, wq = w + g.pad - s;
-        if (hq >= 0 && wq >= 0 && hq % g.stride == 0 && wq % g.stride == 0) {
-          int ho = hq / g.stride, wo = wq / g.stride;
-          if (ho < g.Ho && wo < g.Wo)
-            v = b2f(dy[(((long)n * g.Ho + ho) * g.Wo + wo) * g.K + kk]);
-        }
+  for (int j = 0; j < 8; ++j) {
+    int rr = red + j;
+    float v = 0.f;
+    if (mvalid && rr < RSK) {
+      int kk = rr % g.K;
+      int rs = rr / g.K;
+      int r = rs / g.S, s = rs % g.S;
+      int hq = h + g.pad - r, wq = w + g.pad - s;
+      if (hq >= 0 && wq >= 0 && hq % g.stride == 0 && wq % g.stride == 0) {
+        int ho = hq / g.stride, wo = wq / g.stride;
+        if (ho < g.Ho && wo < g.Wo)
+          v = b2f(dy[(((long)n * g.Ho + ho) * g.Wo + wo) * g.K + kk]);
       }
-      dst[j] = (__bf16)v;
     }
+    dst[j] = (__bf16)v;
   }
 }
 
@@ -158,13 +107,10 @@ __device__ __forceinline__ void row8(const bf16* g, long rows, long cols,
   }
 }
 
-// ---- fwd & dgrad share the main loop (template on the A-stager) ----------
+// ---- fwd & dgrad main loop ----------------------------------------------
 // 128x64 block tile, BK=64, 8 waves (4x2) of 32x32, 512 threads.
 
-constexpr int CBM = 128, CBN = 64, CBK = 64;
-constexpr int CLDK = CBK + 8;
-
-template <int MODE>  // 0 = fwd, 1 = dgrad
+template <int MODE, bool FAST>  // MODE: 0 = fwd, 1 = dgrad
 __global__ __launch_bounds__(512) void conv_mm_kernel(
     const bf16* __restrict__ Asrc, const bf16* __restrict__ B,
     const float* __restrict__ bias, bf16* __restrict__ out, Geom g,
@@ -176,31 +122,88 @@ __global__ __launch_bounds__(512) void conv_mm_kernel(
   const long n0 = (long)blockIdx.y * CBN;
   const int t = threadIdx.x;
   const int lane = t & 63;
-  const int wave = t >> 6;          // 8 waves: wr in 0..3, wc in 0..1
+  const int wave = t >> 6;
   const int wr = wave >> 1, wc = wave & 1;
-  // A staging: 512 thr x 2 vec8 = 128x64; B staging: 512 thr x 1 vec8 = 64x64
-  const int lra = t >> 2;            // 0..127
-  const int lka = (t & 3) * 8;       // 0..24
-  const int lrb = t >> 3;            // 0..63
-  const int lkb = (t & 7) * 8;       // 0..56
+  const int lra = t >> 2;            // A staging row 0..127
+  const int lka = (t & 3) * 8;       // A staging col {0,8,16,24}
+  const int lrb = t >> 3;            // B staging row 0..63
+  const int lkb = (t & 7) * 8;       // B staging col {0..56}
+
+  // hoisted per-thread A-row pixel decomposition
+  const long m = m0 + lra;
+  const bool mvalid = m < M;
+  int pn = 0, ph = 0, pw = 0;  // MODE0: (n, ho, wo); MODE1: (n, h, w)
+  if (mvalid) {
+    const int HW = MODE == 0 ? g.Ho * g.Wo : g.H * g.W;
+    const int WW = MODE == 0 ? g.Wo : g.W;
+    pn = (int)(m / HW);
+    int rem = (int)(m % HW);
+    ph = rem / WW;
+    pw = rem % WW;
+  }
+  // reduction cursors for the two staged vec8 positions (lka, lka+32).
+  // inner = c (MODE0) or k (MODE1); rs = r*S + s.
+  const int INNER = MODE == 0 ? g.C : g.K;
+  int cur_i[2], cur_rs[2];
+#pragma unroll
+  for (int p = 0; p < 2; ++p) {
+    int red = lka + 32 * p;
+    cur_i[p] = FAST ? red % INNER : red;  // slow path keeps raw red
+    cur_rs[p] = FAST ? red / INNER : 0;
+  }
 
   f32x4 acc[2][2] = {};
 
   for (long k0 = 0; k0 < RED; k0 += CBK) {
-    __bf16 ra0[8], ra1[8], rb[8];
-    if (MODE == 0) {
-      stage_im2col8(Asrc, g, m0 + lra, (int)(k0 + lka), ra0);
-      stage_im2col8(Asrc, g, m0 + lra, (int)(k0 + lka + 32), ra1);
-    } else {
-      stage_dgrad8(Asrc, g, m0 + lra, (int)(k0 + lka), ra0);
-      stage_dgrad8(Asrc, g, m0 + lra, (int)(k0 + lka + 32), ra1);
+    __bf16 ra[2][8], rb[8];
+#pragma unroll
+    for (int p = 0; p < 2; ++p) {
+      if (FAST) {
+        const int r = cur_rs[p] / g.S, s = cur_rs[p] % g.S;
+        bool valid;
+        long off;
+        if (MODE == 0) {
+          int h = ph * g.stride - g.pad + r;
+          int w = pw * g.stride - g.pad + s;
+          valid = mvalid && (unsigned)h < (unsigned)g.H &&
+                  (unsigned)w < (unsigned)g.W;
+          off = (((long)pn * g.H + h) * g.W + w) * g.C + cur_i[p];
+        } else {
+          int hq = ph + g.pad - r, wq = pw + g.pad - s;
+          bool div_ok = (g.stride == 1) || (((hq | wq) & 1) == 0);
+          int ho = g.stride == 2 ? (hq >> 1) : hq;
+          int wo = g.stride == 2 ? (wq >> 1) : wq;
+          valid = mvalid && hq >= 0 && wq >= 0 && div_ok && ho < g.Ho &&
+                  wo < g.Wo;
+          off = (((long)pn * g.Ho + ho) * g.Wo + wo) * g.K + cur_i[p];
+        }
+        if (valid) {
+          *reinterpret_cast<s16x8*>(ra[p]) =
+              *reinterpret_cast<const s16x8*>(Asrc + off);
+        } else {
+          zero8(ra[p]);
+        }
+        // advance cursor by CBK elements along the inner dim
+        cur_i[p] += CBK;
+        while (cur_i[p] >= INNER) {
+          cur_i[p] -= INNER;
+          cur_rs[p] += 1;
+        }
+      } else {
+        int red = (int)k0 + lka + 32 * p;
+        if (MODE == 0) {
+          stage_im2col_slow(Asrc, g, mvalid, pn, ph, pw, red, ra[p]);
+        } else {
+          stage_dgrad_slow(Asrc, g, mvalid, pn, ph, pw, red, ra[p]);
+        }
+      }
     }
     row8(B, N, RED, n0 + lrb, k0 + lkb, rb);
     __syncthreads();
     *reinterpret_cast<bf16x8*>(&sA[lra * CLDK + lka]) =
-        *reinterpret_cast<bf16x8*>(ra0);
+        *reinterpret_cast<bf16x8*>(ra[0]);
     *reinterpret_cast<bf16x8*>(&sA[lra * CLDK + lka + 32]) =
-        *reinterpret_cast<bf16x8*>(ra1);
+        *reinterpret_cast<bf16x8*>(ra[1]);
     *reinterpret_cast<bf16x8*>(&sB[lrb * CLDK + lkb]) =
         *reinterpret_cast<bf16x8*>(rb);
     __syncthreads();
@@ -242,65 +245,135 @@ __global__ __launch_bounds__(512) void conv_mm_kernel(
 }
 
 // ---- wgrad: dw[k, rsc] = sum_np dy[np,k] * im2col[np,rsc]  (TN) ----------
+// 64x64 tile over (k, rsc), WBK=64 deep in np, transposed LDS staging.
 
-constexpr int WLDK = BK + 8;  // padded np-stride for transposed LDS tiles
+constexpr int WBM = 64, WBN = 64, WBK = 64;
+constexpr int WLDN = WBK + 8;  // padded np-stride for transposed LDS tiles
 
+template <bool FAST>  // FAST: K % 8 == 0 && C % 8 == 0
 __global__ __launch_bounds__(256) void conv_wgrad_kernel(
     const bf16* __restrict__ x, const bf16* __restrict__ dy,
     float* __restrict__ dw, Geom g, long NP, long npslice) {
-  // sA[k(64)][np(32)], sB[rsc(64)][np(32)] — transposed staging
-  __shared__ __bf16 sA[BM * WLDK];
-  __shared__ __bf16 sB[BN * WLDK];
+  // sA[k(64)][np(64)], sB[rsc(64)][np(64)] — transposed staging
+  __shared__ __bf16 sA[WBM * WLDN];
+  __shared__ __bf16 sB[WBN * WLDN];
 
-  const long k0c = (long)blockIdx.x * BM;   // out-channel tile
-  const long n0 = (long)blockIdx.y * BN;    // rsc tile
+  const long k0c = (long)blockIdx.x * WBM;   // out-channel tile
+  const long n0 = (long)blockIdx.y * WBN;    // rsc tile
   const long np0 = (long)blockIdx.z * npslice;
   const long np1 = min(np0 + npslice, NP);
-  const long RED = g.R * g.S * g.C;
+  const long RED = (long)g.R * g.S * g.C;
 
   const int t = threadIdx.x;
   const int lane = t & 63;
   const int wave = t >> 6;
   const int wr = wave >> 1, wc = wave & 1;
-  // staging map: 256 threads = 32 np-rows x 8 col-chunks of 8
-  const int snp = t >> 3;          // 0..31 np row
-  const int scol = (t & 7) * 8;    // 0..56 col chunk
+  // staging map: 256 threads = 32 np-rows x 8 col-chunks of 8; each thread
+  // stages np rows {snp, snp+32}.
+  const int snp = t >> 3;          // 0..31
+  const int scol = (t & 7) * 8;    // 0..56
+
+  // hoisted B-column decomposition: (r, s, c) is FIXED per thread
+  int fb_r = 0, fb_s = 0, fb_c = 0;
+  bool bcol_ok = (n0 + scol) < RED;
+  if (bcol_ok) {
+    int rsc = (int)(n0 + scol);
+    fb_c = rsc % g.C;
+    int rs = rsc / g.C;
+    fb_r = rs / g.S;
+    fb_s = rs % g.S;
+  }
+  const bool b_vec_ok = FAST && bcol_ok && (fb_c + 8 <= g.C);
+
+  // incremental pixel cursors for the two staged np rows
+  int pn[2], pho[2], pwo[2];
+  bool pv[2];
+#pragma unroll
+  for (int q = 0; q < 2; ++q) {
+    long m = np0 + snp + 32 * q;
+    pv[q] = m < np1;
+    long mm = pv[q] ? m : 0;
+    pn[q] = (int)(mm / ((long)g.Ho * g.Wo));
+    int rem = (int)(mm % ((long)g.Ho * g.Wo));
+    pho[q] = rem / g.Wo;
+    pwo[q] = rem % g.Wo;
+  }
 
   f32x4 acc[2][2] = {};
 
-  for (long p0 = np0; p0 < np1; p0 += BK) {
-    __bf16 ra[8], rb[8];
-    long m = p0 + snp;
-    // A: dy[np, k] chunk (vec8 along k), transposed into sA[k][np]
-    if (m < np1) row8(dy, NP, g.K, m, k0c + scol, ra);
-    else {
+  for (long p0 = np0; p0 < np1; p0 += WBK) {
+    __bf16 ra[2][8], rb[2][8];
 #pragma unroll
-      for (int j = 0; j < 8; ++j) ra[j] = (__bf16)0.f;
-    }
-    // B: im2col[np, rsc] chunk, transposed into sB[rsc][np]
-    if (m < np1) stage_im2col8(x, g, m, (int)(n0 + scol), rb);
-    else {
-#pragma unroll
-      for (int j = 0; j < 8; ++j) rb[j] = (__bf16)0.f;
+    for (int q = 0; q < 2; ++q) {
+      const long m = p0 + snp + 32 * q;
+      const bool v = m < np1;
+      // A: dy[np, k] chunk (vec8 along k)
+      if (!v) {
+        zero8(ra[q]);
+      } else if (FAST && k0c + scol + 8 <= g.K) {
+        *reinterpret_cast<s16x8*>(ra[q]) = *reinterpret_cast<const s16x8*>(
+            dy + m * g.K + k0c + scol);
+      } else {
+        row8(dy, NP, g.K, m, k0c + scol, ra[q]);
+      }
+      // B: im2col[np, rsc] chunk at the thread's fixed (r,s,c)
+      if (!v || !bcol_ok) {
+        zero8(rb[q]);
+      } else if (FAST) {
+        int h = pho[q] * g.stride - g.pad + fb_r;
+        int w = pwo[q] * g.stride - g.pad + fb_s;
+        bool valid = (unsigned)h < (unsigned)g.H &&
+                     (unsigned)w < (unsigned)g.W && b_vec_ok;
+        if (valid) {
+          long off = (((long)pn[q] * g.H + h) * g.W + w) * g.C + fb_c;
+          *reinterpret_cast<s16x8*>(rb[q]) =
+              *reinterpret_cast<const s16x8*>(x + off);
+        } else {
+          zero8(rb[q]);
+        }
+      } else {
+        // generic: columns may cross (r,s) boundaries -> per element
+        stage_im2col_slow(x, g, true, pn[q], pho[q], pwo[q],
+                          (int)(n0 + scol), rb[q]);
+      }
+      // advance pixel cursor by WBK rows
+      if (pv[q]) {
+        pwo[q] += WBK;
+        while (pwo[q] >= g.Wo) {
+          pwo[q] -= g.Wo;
+          pho[q] += 1;
+        }
+        while (pho[q] >= g.Ho) {
+          pho[q] -= g.Ho;
+          pn[q] += 1;
+        }
+      }
     }
     __syncthreads();
 #pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      sA[(scol + j) * WLDK + snp] = ra[j];
-      sB[(scol + j) * WLDK + snp] = rb[j];
+    for (int q = 0; q < 2; ++q) {
+      const int npl = snp + 32 * q;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        sA[(scol + j) * WLDN + npl] = ra[q][j];
+        sB[(scol + j) * WLDN + npl] = rb[q][j];
+      }
     }
     __syncthreads();
 
-    const int kf = (lane >> 4) * 8;
     const int ml = lane & 15;
 #pragma unroll
-    for (int fi = 0; fi < 2; ++fi) {
-      bf16x8 af = lds8(&sA[(wr * 32 + fi * 16 + ml) * WLDK + kf]);
+    for (int ks = 0; ks < 2; ++ks) {
+      const int kf = ks * 32 + (lane >> 4) * 8;
 #pragma unroll
-      for (int fj = 0; fj < 2; ++fj) {
-        bf16x8 bfr = lds8(&sB[(wc * 32 + fj * 16 + ml) * WLDK + kf]);
-        acc[fi][fj] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            af, bfr, acc[fi][fj], 0, 0, 0);
+      for (int fi = 0; fi < 2; ++fi) {
+        bf16x8 af = lds8(&sA[(wr * 32 + fi * 16 + ml) * WLDN + kf]);
+#pragma unroll
+        for (int fj = 0; fj < 2; ++fj) {
+          bf16x8 bfr = lds8(&sB[(wc * 32 + fj * 16 + ml) * WLDN + kf]);
+          acc[fi][fj] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af, bfr, acc[fi][fj], 0, 0, 0);
+        }
       }
     }
   }
@@ -350,8 +423,11 @@ torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor bias,
   long M = (long)g.N * g.Ho * g.Wo;
   long RED = (long)g.R * g.S * g.C;
   bool has_bias = bias.numel() > 0;
+  bool fast = (g.C % 8 == 0);
   dim3 grid(ceil_div(M, conv::CBM), ceil_div(g.K, conv::CBN));
-  conv::conv_mm_kernel<0><<<grid, 512, 0, cur_stream()>>>(
+  auto* fn = fast ? conv::conv_mm_kernel<0, true>
+                  : conv::conv_mm_kernel<0, false>;
+  fn<<<grid, 512, 0, cur_stream()>>>(
       (const bf16*)x.data_ptr(), (const bf16*)w.data_ptr(),
       has_bias ? bias.data_ptr<float>() : nullptr, (bf16*)y.data_ptr(), g,
       M, g.K, RED, has_bias ? 1 : 0);
@@ -372,8 +448,11 @@ torch::Tensor conv2d_dgrad(torch::Tensor dy, torch::Tensor wt, long stride,
   auto dx = torch::empty({g.N, g.H, g.W, g.C}, dy.options());
   long M = (long)g.N * g.H * g.W;
   long RED = (long)R * S * K;
+  bool fast = (g.K % 8 == 0) && (g.stride <= 2);
   dim3 grid(ceil_div(M, conv::CBM), ceil_div(C, conv::CBN));
-  conv::conv_mm_kernel<1><<<grid, 512, 0, cur_stream()>>>(
+  auto* fn = fast ? conv::conv_mm_kernel<1, true>
+                  : conv::conv_mm_kernel<1, false>;
+  fn<<<grid, 512, 0, cur_stream()>>>(
       (const bf16*)dy.data_ptr(), (const bf16*)wt.data_ptr(), nullptr,
       (bf16*)dx.data_ptr(), g, M, C, RED, 0);
   return dx;
@@ -391,15 +470,18 @@ torch::Tensor conv2d_wgrad(torch::Tensor x, torch::Tensor dy, long R, long S,
   auto dw = torch::zeros({(long)g.K, RED}, x.options().dtype(torch::kFloat32));
   // split the NP reduction across blocks for parallelism; fp32 atomics
   long target_blocks = 1024;
-  long tiles = (long)ceil_div(g.K, conv::BM) * ceil_div(RED, conv::BN);
+  long tiles = (long)ceil_div(g.K, conv::WBM) * ceil_div(RED, conv::WBN);
   long zsplit =
       std::max(1L, std::min(512L, target_blocks / std::max(tiles, 1L)));
   long npslice = (NP + zsplit - 1) / zsplit;
-  npslice = ((npslice + conv::BK - 1) / conv::BK) * conv::BK;
+  npslice = ((npslice + conv::WBK - 1) / conv::WBK) * conv::WBK;
   zsplit = (NP + npslice - 1) / npslice;
-  dim3 grid(ceil_div(g.K, conv::BM), ceil_div(RED, conv::BN),
+  bool fast = (g.K % 8 == 0) && (g.C % 8 == 0);
+  dim3 grid(ceil_div(g.K, conv::WBM), ceil_div(RED, conv::WBN),
             (unsigned)zsplit);
-  conv::conv_wgrad_kernel<<<grid, 256, 0, cur_stream()>>>(
+  auto* fn = fast ? conv::conv_wgrad_kernel<true>
+                  : conv::conv_wgrad_kernel<false>;
+  fn<<<grid, 256, 0, cur_stream()>>>(
       (const bf16*)x.data_ptr(), (const bf16*)dy.data_ptr(),
       dw.data_ptr<float>(), g, NP, npslice);
   return dw.view({(long)g.K, R, S, (long)g.C});

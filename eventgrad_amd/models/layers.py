@@ -57,10 +57,19 @@ class BatchNorm2d(nn.Module):
         self.register_buffer("running_var", torch.ones(num_features))
         self.register_buffer("num_batches_tracked",
                              torch.zeros(1, dtype=torch.long))
+        self._nbt = 0  # python-side mirror; avoids one GPU kernel per step
+
+    def _save_to_state_dict(self, destination, prefix, keep_vars):
+        self.num_batches_tracked.fill_(self._nbt)
+        super()._save_to_state_dict(destination, prefix, keep_vars)
+
+    def _load_from_state_dict(self, *args, **kwargs):
+        super()._load_from_state_dict(*args, **kwargs)
+        self._nbt = int(self.num_batches_tracked.item())
 
     def forward(self, x, fuse_relu: bool = False):
         if self.training:
-            self.num_batches_tracked += 1
+            self._nbt += 1
         return O.batch_norm(x, self.weight, self.bias, self.running_mean,
                             self.running_var, self.training, self.momentum,
                             self.eps, fuse_relu)
