@@ -245,16 +245,171 @@ __global__ __launch_bounds__(512) void conv_mm_kernel(
 }
 
 // ---- wgrad: dw[k, rsc] = sum_np dy[np,k] * im2col[np,rsc]  (TN) ----------
-// 64x64 tile over (k, rsc), WBK=64 deep in np, transposed LDS staging.
+// 64x64 tile over (k, rsc), WBK=64 deep in np. Both operands need their
+// MFMA fragments transposed relative to the global [np, inner] layout; the
+// FAST kernel stages LINEAR 16-B writes into a [inner/16][np][16] image and
+// reads fragments with ds_read_b64_tr_b16 (gfx950 LDS hardware transpose:
+// each lane gets 4 bf16 strided 16 elements from its own address), so the
+// per-element transposed ds_write storm of the naive version disappears.
 
 constexpr int WBM = 64, WBN = 64, WBK = 64;
-constexpr int WLDN = WBK + 8;  // padded np-stride for transposed LDS tiles
+constexpr int WLDN = WBK + 8;  // padded np-stride (generic fallback kernel)
 
-template <bool FAST>  // FAST: K % 8 == 0 && C % 8 == 0
+typedef __bf16 bf16x4t __attribute__((ext_vector_type(4)));
+typedef bf16x4t __attribute__((address_space(3)))* lds_b64p;
+
+// two tr16 reads -> one 8-elem fragment: lane l receives image elements
+// [np0 + (l>>4)*8 + j][l&15] of a [np][16] bf16 image starting at `base`.
+__device__ __forceinline__ bf16x8 tr16_frag(const __bf16* base, int np0,
+                                            int lane) {
+  const __bf16* p = base + (np0 + ((lane >> 4) << 3)) * 16 + (lane & 15);
+  auto p3 = (lds_b64p)(__bf16 __attribute__((address_space(3)))*)p;
+  bf16x4t v0 = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(p3);
+  bf16x4t v1 = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(p3 + 16);
+  bf16x8 r;
+#pragma unroll
+  for (int j = 0; j < 4; ++j) {
+    r[j] = v0[j];
+    r[4 + j] = v1[j];
+  }
+  return r;
+}
+
+__global__ __launch_bounds__(256) void conv_wgrad_fast_kernel(
+    const bf16* __restrict__ x, const bf16* __restrict__ dy,
+    float* __restrict__ dw, Geom g, long NP, long npslice) {
+  // images: [inner16-block][np(64)][16] per operand (4 blocks of 64 k/rsc)
+  __shared__ __bf16 sA[4 * WBK * 16];
+  __shared__ __bf16 sB[4 * WBK * 16];
+
+  const long k0c = (long)blockIdx.x * WBM;
+  const long n0 = (long)blockIdx.y * WBN;
+  const long np0 = (long)blockIdx.z * npslice;
+  const long np1 = min(np0 + npslice, NP);
+  const long RED = (long)g.R * g.S * g.C;
+
+  const int t = threadIdx.x;
+  const int lane = t & 63;
+  const int wave = t >> 6;
+  const int wr = wave >> 1, wc = wave & 1;
+  const int snp = t >> 3;          // 0..31 (stages rows snp, snp+32)
+  const int scol = (t & 7) * 8;    // 0..56
+  const int simg = (scol >> 4) * (WBK * 16) + (scol & 15);
+
+  // fixed per-thread B column (r, s, c)
+  int fb_r = 0, fb_s = 0, fb_c = 0;
+  const bool bcol_ok = (n0 + scol) < RED;
+  if (bcol_ok) {
+    int rsc = (int)(n0 + scol);
+    fb_c = rsc % g.C;
+    int rs = rsc / g.C;
+    fb_r = rs / g.S;
+    fb_s = rs % g.S;
+  }
+  const bool a_ok = k0c + scol + 8 <= g.K;
+
+  // incremental pixel cursors for rows snp and snp+32
+  int pn[2], pho[2], pwo[2];
+#pragma unroll
+  for (int q = 0; q < 2; ++q) {
+    long m = np0 + snp + 32 * q;
+    long mm = m < NP ? m : 0;
+    pn[q] = (int)(mm / ((long)g.Ho * g.Wo));
+    int rem = (int)(mm % ((long)g.Ho * g.Wo));
+    pho[q] = rem / g.Wo;
+    pwo[q] = rem % g.Wo;
+  }
+
+  f32x4 acc[2][2] = {};
+
+  for (long p0 = np0; p0 < np1; p0 += WBK) {
+    __bf16 ra[2][8], rb[2][8];
+#pragma unroll
+    for (int q = 0; q < 2; ++q) {
+      const long m = p0 + snp + 32 * q;
+      const bool v = m < np1;
+      if (v && a_ok) {
+        *reinterpret_cast<s16x8*>(ra[q]) = *reinterpret_cast<const s16x8*>(
+            dy + m * g.K + k0c + scol);
+      } else {
+        zero8(ra[q]);
+      }
+      if (v && bcol_ok) {
+        int h = pho[q] * g.stride - g.pad + fb_r;
+        int w = pwo[q] * g.stride - g.pad + fb_s;
+        if ((unsigned)h < (unsigned)g.H && (unsigned)w < (unsigned)g.W) {
+          long off = (((long)pn[q] * g.H + h) * g.W + w) * g.C + fb_c;
+          *reinterpret_cast<s16x8*>(rb[q]) =
+              *reinterpret_cast<const s16x8*>(x + off);
+        } else {
+          zero8(rb[q]);
+        }
+      } else {
+        zero8(rb[q]);
+      }
+      if (v) {
+        pwo[q] += WBK;
+        while (pwo[q] >= g.Wo) {
+          pwo[q] -= g.Wo;
+          pho[q] += 1;
+        }
+        while (pho[q] >= g.Ho) {
+          pho[q] -= g.Ho;
+          pn[q] += 1;
+        }
+      }
+    }
+    __syncthreads();
+#pragma unroll
+    for (int q = 0; q < 2; ++q) {
+      const int npl = snp + 32 * q;
+      *reinterpret_cast<bf16x8*>(&sA[simg + npl * 16]) =
+          *reinterpret_cast<bf16x8*>(ra[q]);
+      *reinterpret_cast<bf16x8*>(&sB[simg + npl * 16]) =
+          *reinterpret_cast<bf16x8*>(rb[q]);
+    }
+    __syncthreads();
+
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks) {
+#pragma unroll
+      for (int fi = 0; fi < 2; ++fi) {
+        bf16x8 af =
+            tr16_frag(&sA[(wr * 2 + fi) * (WBK * 16)], ks * 32, lane);
+#pragma unroll
+        for (int fj = 0; fj < 2; ++fj) {
+          bf16x8 bfr =
+              tr16_frag(&sB[(wc * 2 + fj) * (WBK * 16)], ks * 32, lane);
+          acc[fi][fj] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af, bfr, acc[fi][fj], 0, 0, 0);
+        }
+      }
+    }
+  }
+
+  const int cn = lane & 15;
+  const int cm = (lane >> 4) * 4;
+#pragma unroll
+  for (int fi = 0; fi < 2; ++fi) {
+#pragma unroll
+    for (int fj = 0; fj < 2; ++fj) {
+      long nn = n0 + wc * 32 + fj * 16 + cn;
+      if (nn >= RED) continue;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        long kk = k0c + wr * 32 + fi * 16 + cm + r;
+        if (kk >= g.K) continue;
+        atomicAdd(&dw[kk * RED + nn], acc[fi][fj][r]);
+      }
+    }
+  }
+}
+
+template <bool FAST>  // generic fallback (any C/K); WBK_GEN=32 transposed
 __global__ __launch_bounds__(256) void conv_wgrad_kernel(
     const bf16* __restrict__ x, const bf16* __restrict__ dy,
     float* __restrict__ dw, Geom g, long NP, long npslice) {
-  // sA[k(64)][np(64)], sB[rsc(64)][np(64)] — transposed staging
+  // sA[k(64)][np(32)], sB[rsc(64)][np(32)] — transposed staging
   __shared__ __bf16 sA[WBM * WLDN];
   __shared__ __bf16 sB[WBN * WLDN];
 
@@ -479,11 +634,15 @@ torch::Tensor conv2d_wgrad(torch::Tensor x, torch::Tensor dy, long R, long S,
   bool fast = (g.K % 8 == 0) && (g.C % 8 == 0);
   dim3 grid(ceil_div(g.K, conv::WBM), ceil_div(RED, conv::WBN),
             (unsigned)zsplit);
-  auto* fn = fast ? conv::conv_wgrad_kernel<true>
-                  : conv::conv_wgrad_kernel<false>;
-  fn<<<grid, 256, 0, cur_stream()>>>(
-      (const bf16*)x.data_ptr(), (const bf16*)dy.data_ptr(),
-      dw.data_ptr<float>(), g, NP, npslice);
+  if (fast) {
+    conv::conv_wgrad_fast_kernel<<<grid, 256, 0, cur_stream()>>>(
+        (const bf16*)x.data_ptr(), (const bf16*)dy.data_ptr(),
+        dw.data_ptr<float>(), g, NP, npslice);
+  } else {
+    conv::conv_wgrad_kernel<false><<<grid, 256, 0, cur_stream()>>>(
+        (const bf16*)x.data_ptr(), (const bf16*)dy.data_ptr(),
+        dw.data_ptr<float>(), g, NP, npslice);
+  }
   return dw.view({(long)g.K, R, S, (long)g.C});
 }
 
