@@ -54,6 +54,7 @@ torch::Tensor logsoftmax_nll_bwd(torch::Tensor, torch::Tensor, double);
 // topk.hip
 std::vector<torch::Tensor> topk_absdiff(torch::Tensor, torch::Tensor, long);
 void scatter_update(torch::Tensor, torch::Tensor, torch::Tensor);
+torch::Tensor tr16_probe(long);
 }  // namespace eg
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -86,4 +87,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("logsoftmax_nll_bwd", &eg::logsoftmax_nll_bwd);
   m.def("topk_absdiff", &eg::topk_absdiff);
   m.def("scatter_update", &eg::scatter_update);
+  m.def("tr16_probe", &eg::tr16_probe);
 }

@@ -260,9 +260,16 @@ typedef bf16x4t __attribute__((address_space(3)))* lds_b64p;
 
 // two tr16 reads -> one 8-elem fragment: lane l receives image elements
 // [np0 + (l>>4)*8 + j][l&15] of a [np][16] bf16 image starting at `base`.
+//
+// ds_read_b64_tr_b16 semantics (measured with tr16_probe, all 4 modes):
+// within a 16-lane group, lane i receives in reg j (j=0..3) the 16-bit
+// element at ALIGN8(byte addr supplied by lane 4j + (i>>2)) + 2*(i&3).
+// So to deliver img[np0+8g+j][i] to lane i of group g, the lane at group
+// coords (a = (l>>2)&3, b = l&3) must SUPPLY elem (np0 + 8g + a)*16 + 4b.
 __device__ __forceinline__ bf16x8 tr16_frag(const __bf16* base, int np0,
                                             int lane) {
-  const __bf16* p = base + (np0 + ((lane >> 4) << 3)) * 16 + (lane & 15);
+  const __bf16* p = base +
+      (np0 + ((lane >> 4) << 3) + ((lane >> 2) & 3)) * 16 + (lane & 3) * 4;
   auto p3 = (lds_b64p)(__bf16 __attribute__((address_space(3)))*)p;
   bf16x4t v0 = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(p3);
   bf16x4t v1 = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(p3 + 16);
@@ -551,7 +558,37 @@ __global__ __launch_bounds__(256) void conv_wgrad_kernel(
   }
 }
 
+// debug probe: LDS filled with element indices; dump what each lane's
+// tr16_frag (as used by wgrad) actually receives.
+__global__ void tr16_probe_kernel(float* out, int mode) {
+  __shared__ __bf16 s[4096];
+  for (int i = threadIdx.x; i < 4096; i += blockDim.x)
+    s[i] = (__bf16)(float)i;
+  __syncthreads();
+  if (threadIdx.x < 64) {
+    int lane = threadIdx.x;
+    long elem = 0;
+    if (mode == 0) elem = ((lane >> 4) << 3) * 16 + (lane & 15);
+    if (mode == 1) elem = lane * 16;
+    if (mode == 2) elem = 0;
+    if (mode == 3) elem = lane * 4;
+    const __bf16* p = &s[elem];
+    auto p3 = (lds_b64p)(__bf16 __attribute__((address_space(3)))*)p;
+    bf16x4t v0 = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(p3);
+#pragma unroll
+    for (int j = 0; j < 4; ++j) out[lane * 4 + j] = (float)v0[j];
+  }
+}
+
 }  // namespace conv
+
+torch::Tensor tr16_probe(long mode) {
+  auto out = torch::zeros({64, 4},
+                          torch::dtype(torch::kFloat32).device(torch::kCUDA));
+  conv::tr16_probe_kernel<<<1, 64, 0, cur_stream()>>>(out.data_ptr<float>(),
+                                                      (int)mode);
+  return out;
+}
 
 static conv::Geom make_geom(const torch::Tensor& x, int K, int R, int S,
                             long stride, long pad) {
