@@ -26,10 +26,13 @@ class SyntheticImages:
     """
 
     def __init__(self, shape: Tuple[int, int, int], n: int,
-                 num_classes: int = 10, noise: float = 0.5, seed: int = 1234):
+                 num_classes: int = 10, noise: float = 0.5, seed: int = 1234,
+                 prototype_seed: int = 977):
         self.shape, self.n, self.num_classes = shape, n, num_classes
         self.noise = noise
-        g = torch.Generator().manual_seed(seed)
+        # prototypes are shared between train/test splits (same
+        # prototype_seed); only the per-index noise differs via `seed`.
+        g = torch.Generator().manual_seed(prototype_seed)
         self.prototypes = torch.randn((num_classes,) + shape, generator=g)
         self.seed = seed
 
