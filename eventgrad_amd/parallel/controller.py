@@ -99,3 +99,86 @@ class TriggerController:
         self.last_sent_iter = d["last_sent_iter"].astype(np.float32).copy()
         self.slopes = d["slopes"].astype(np.float32).copy()
         self.num_events = int(d["num_events"])
+
+
+class GpuTriggerController:
+    """Device-resident controller: state lives in GPU tensors and the whole
+    trigger evaluation + threshold adaptation runs in the HIP kernel
+    (csrc/engine.hip trigger_update), fused with the SGD-step kernel's
+    norm output — only the fire mask (sz bytes) crosses to the host, to
+    size the RCCL payloads. Kernel parity with TriggerController is
+    covered by tests/test_gpu_numerics.py::test_trigger_update_matches_host_controller.
+    """
+
+    def __init__(self, sz: int, device, adaptive: bool, horizon: float,
+                 constant: float, sent_history: int = 2,
+                 initial_comm_passes: int = 30, always_fire: bool = False):
+        import torch
+
+        self.sz = sz
+        self.device = device
+        self.adaptive = adaptive
+        self.horizon = float(horizon)
+        self.constant = float(constant)
+        self.sent_history = sent_history
+        self.initial_comm_passes = initial_comm_passes
+        self.always_fire = always_fire
+
+        f32 = dict(dtype=torch.float32, device=device)
+        self.thres = torch.zeros(sz, **f32)
+        self.last_sent_norm = torch.zeros(sz, **f32)
+        self.last_sent_iter = torch.zeros(sz, **f32)
+        self.slopes = torch.zeros(sz * sent_history, **f32)
+        self._num_events = torch.zeros(1, dtype=torch.int32, device=device)
+        self.last_norms = None   # device sqnorms of the last step (for trace)
+        self.last_fired = None
+
+    @property
+    def num_events(self) -> int:
+        return int(self._num_events.item())
+
+    @num_events.setter
+    def num_events(self, v: int):
+        self._num_events.fill_(int(v))
+
+    def step_device(self, norms_sq, pass_num: int):
+        """norms_sq: device fp32[sz] (squared L2 norms). Returns np bool[sz]."""
+        from ..ops.backend import native
+
+        mask = native().trigger_update(
+            norms_sq, self.thres, self.last_sent_norm, self.last_sent_iter,
+            self.slopes, self._num_events, pass_num, self.adaptive,
+            self.horizon, self.constant, self.initial_comm_passes,
+            self.always_fire)
+        self.last_norms = norms_sq
+        fired = mask.cpu().numpy().astype(bool)
+        self.last_fired = fired
+        return fired
+
+    def trace_values(self):
+        """(norms, thres) as numpy — only used when tracing is enabled."""
+        import numpy as _np
+        norms = _np.sqrt(self.last_norms.cpu().numpy()) \
+            if self.last_norms is not None else _np.zeros(self.sz, _np.float32)
+        return norms, self.thres.cpu().numpy()
+
+    # -- checkpointing (same dict format as the host controller) ----------
+    def state_dict(self) -> dict:
+        return {
+            "thres": self.thres.cpu().numpy(),
+            "last_sent_norm": self.last_sent_norm.cpu().numpy(),
+            "last_sent_iter": self.last_sent_iter.cpu().numpy(),
+            "slopes": self.slopes.cpu().numpy().reshape(
+                self.sz, self.sent_history),
+            "num_events": self.num_events,
+        }
+
+    def load_state_dict(self, d: dict) -> None:
+        import torch
+
+        for name in ("thres", "last_sent_norm", "last_sent_iter"):
+            getattr(self, name).copy_(
+                torch.from_numpy(np.asarray(d[name], dtype=np.float32)))
+        self.slopes.copy_(torch.from_numpy(
+            np.asarray(d["slopes"], dtype=np.float32).reshape(-1)))
+        self.num_events = int(d["num_events"])

@@ -138,8 +138,24 @@ class CommEngine:
         self.device = device
         self.K = _kernels(device)
         self.tracer = tracer
-        self.num_events = 0
         self.pass_num = 0
+
+    # num_events lives in the controller when one exists (device-resident
+    # on GPU); engines without a trigger (serial/cent) keep a plain counter.
+    @property
+    def num_events(self) -> int:
+        ctrl = getattr(self, "ctrl", None)
+        if ctrl is not None:
+            return ctrl.num_events
+        return getattr(self, "_num_events", 0)
+
+    @num_events.setter
+    def num_events(self, v: int) -> None:
+        ctrl = getattr(self, "ctrl", None)
+        if ctrl is not None:
+            ctrl.num_events = v
+        else:
+            self._num_events = v
 
     # lifecycle
     def begin_pass(self, pass_num: int) -> None:
@@ -201,11 +217,15 @@ class GossipEngine(CommEngine):
     def __init__(self, space, cfg, rank, world, device, tracer=None):
         super().__init__(space, cfg, rank, world, device, tracer)
         t = cfg.trigger
-        self.ctrl = TriggerController(
-            space.sz, adaptive=t.adaptive, horizon=t.horizon,
-            constant=t.constant, sent_history=t.sent_history,
-            initial_comm_passes=t.initial_comm_passes,
-            always_fire=(cfg.mode == "decent"))
+        kw = dict(adaptive=t.adaptive, horizon=t.horizon,
+                  constant=t.constant, sent_history=t.sent_history,
+                  initial_comm_passes=t.initial_comm_passes,
+                  always_fire=(cfg.mode == "decent"))
+        if device.type == "cuda":
+            from .controller import GpuTriggerController
+            self.ctrl = GpuTriggerController(space.sz, device, **kw)
+        else:
+            self.ctrl = TriggerController(space.sz, **kw)
         self.transport = (RingTransport(rank, world, device)
                           if world > 1 else None)
         self.inbox_left = space.new_like()
@@ -232,11 +252,20 @@ class GossipEngine(CommEngine):
         self.pass_num = pass_num
         if self.world <= 1:
             return
-        norms = self._norms()
-        fire = self.ctrl.step(norms, pass_num)
-        self.num_events = self.ctrl.num_events
-        if self.tracer is not None:
-            self.tracer.send_line(norms, self.ctrl.thres, fire)
+        if self.device.type == "cuda":
+            # device-resident controller: only the mask crosses to host
+            sq = getattr(self, "_last_norms_sq", None)
+            if sq is None:
+                sq = self.K.sqnorms(self.space, self.space.param)
+            fire = self.ctrl.step_device(sq, pass_num)
+            if self.tracer is not None:
+                norms, thres = self.ctrl.trace_values()
+                self.tracer.send_line(norms, thres, fire)
+        else:
+            norms = self._norms()
+            fire = self.ctrl.step(norms, pass_num)
+            if self.tracer is not None:
+                self.tracer.send_line(norms, self.ctrl.thres, fire)
         mask = torch.from_numpy(fire.astype(np.uint8))
         mask_l, mask_r = self.transport.exchange_masks(mask)
         my_fired = [i for i in range(self.space.sz) if fire[i]]
