@@ -99,6 +99,30 @@ __global__ void eval_stats_kernel(const float* __restrict__ running_mean,
   invstd[ch] = rsqrtf(running_var[ch] + eps);
 }
 
+__global__ void norm_kernel_v(const bf16* __restrict__ x,
+                              bf16* __restrict__ y,
+                              const float* __restrict__ mean,
+                              const float* __restrict__ invstd,
+                              const float* __restrict__ gamma,
+                              const float* __restrict__ beta, long n8, int c8,
+                              int relu) {
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (long v = (long)blockIdx.x * blockDim.x + threadIdx.x; v < n8;
+       v += stride) {
+    int c0 = (int)(v % c8) * 8;
+    s16x8 xv = reinterpret_cast<const s16x8*>(x)[v];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      int ch = c0 + j;
+      float f = (b2f(__ushort_as_bfloat16((unsigned short)xv[j])) - mean[ch])
+                * invstd[ch] * gamma[ch] + beta[ch];
+      if (relu && f < 0.f) f = 0.f;
+      xv[j] = (short)__bfloat16_as_ushort(f2b(f));
+    }
+    reinterpret_cast<s16x8*>(y)[v] = xv;
+  }
+}
+
 __global__ void norm_kernel(const bf16* __restrict__ x, bf16* __restrict__ y,
                             const float* __restrict__ mean,
                             const float* __restrict__ invstd,
@@ -197,6 +221,47 @@ __global__ void bwd_stats_kernel(const bf16* __restrict__ dy,
   atomicAdd(&sum_dyx[ch], b);
 }
 
+__global__ void bwd_dx_kernel_v(const bf16* __restrict__ dy,
+                                const bf16* __restrict__ x,
+                                const bf16* __restrict__ y,
+                                const float* __restrict__ mean,
+                                const float* __restrict__ invstd,
+                                const float* __restrict__ gamma,
+                                const float* __restrict__ sum_dy,
+                                const float* __restrict__ sum_dyx,
+                                bf16* __restrict__ dx, long n8, long rows,
+                                int c8, int relu, int training) {
+  const long stride = (long)gridDim.x * blockDim.x;
+  const float inv_n = 1.0f / (float)rows;
+  for (long v = (long)blockIdx.x * blockDim.x + threadIdx.x; v < n8;
+       v += stride) {
+    int c0 = (int)(v % c8) * 8;
+    s16x8 vg = reinterpret_cast<const s16x8*>(dy)[v];
+    s16x8 vx = reinterpret_cast<const s16x8*>(x)[v];
+    s16x8 vy;
+    if (relu) vy = reinterpret_cast<const s16x8*>(y)[v];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      int ch = c0 + j;
+      float gg = b2f(__ushort_as_bfloat16((unsigned short)vg[j]));
+      if (relu && b2f(__ushort_as_bfloat16((unsigned short)vy[j])) <= 0.f)
+        gg = 0.f;
+      float is = invstd[ch];
+      float f;
+      if (training) {
+        float xh = (b2f(__ushort_as_bfloat16((unsigned short)vx[j]))
+                    - mean[ch]) * is;
+        f = gamma[ch] * is *
+            (gg - sum_dy[ch] * inv_n - xh * sum_dyx[ch] * inv_n);
+      } else {
+        f = gamma[ch] * is * gg;
+      }
+      vg[j] = (short)__bfloat16_as_ushort(f2b(f));
+    }
+    reinterpret_cast<s16x8*>(dx)[v] = vg;
+  }
+}
+
 __global__ void bwd_dx_kernel(const bf16* __restrict__ dy,
                               const bf16* __restrict__ x,
                               const bf16* __restrict__ y,
@@ -247,7 +312,7 @@ std::vector<torch::Tensor> bn_fwd(torch::Tensor x, torch::Tensor gamma,
     auto sq = ssq.narrow(0, c, c);
     if (c % 8 == 0 && c <= bn::BN_MAXC) {
       int rpb = 256 / (c / 8);
-      int grid = (int)std::min<long>((rows + rpb - 1) / rpb, 1024L);
+      int grid = (int)std::min<long>((rows + rpb - 1) / rpb, 4096L);
       bn::stats_kernel_v<<<grid, 256, 0, stream>>>(
           (const bf16*)x.data_ptr(), s.data_ptr<float>(),
           sq.data_ptr<float>(), rows, c);
@@ -270,11 +335,21 @@ std::vector<torch::Tensor> bn_fwd(torch::Tensor x, torch::Tensor gamma,
   }
   auto y = torch::empty_like(x);
   long n = x.numel();
-  int grid1 = (int)std::min<long>((n + 255) / 256, 4096L);
-  bn::norm_kernel<<<grid1, 256, 0, stream>>>(
-      (const bf16*)x.data_ptr(), (bf16*)y.data_ptr(), mean.data_ptr<float>(),
-      invstd.data_ptr<float>(), gamma.data_ptr<float>(),
-      beta.data_ptr<float>(), n, c, relu ? 1 : 0);
+  if (c % 8 == 0) {
+    long n8 = n / 8;
+    int grid1 = (int)std::min<long>((n8 + 255) / 256, 4096L);
+    bn::norm_kernel_v<<<grid1, 256, 0, stream>>>(
+        (const bf16*)x.data_ptr(), (bf16*)y.data_ptr(),
+        mean.data_ptr<float>(), invstd.data_ptr<float>(),
+        gamma.data_ptr<float>(), beta.data_ptr<float>(), n8, c / 8,
+        relu ? 1 : 0);
+  } else {
+    int grid1 = (int)std::min<long>((n + 255) / 256, 4096L);
+    bn::norm_kernel<<<grid1, 256, 0, stream>>>(
+        (const bf16*)x.data_ptr(), (bf16*)y.data_ptr(),
+        mean.data_ptr<float>(), invstd.data_ptr<float>(),
+        gamma.data_ptr<float>(), beta.data_ptr<float>(), n, c, relu ? 1 : 0);
+  }
   return {y, mean, invstd};
 }
 
@@ -293,7 +368,7 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor x,
   auto stream = cur_stream();
   if (c % 8 == 0 && c <= bn::BN_MAXC) {
     int rpb = 256 / (c / 8);
-    int grid = (int)std::min<long>((rows + rpb - 1) / rpb, 1024L);
+    int grid = (int)std::min<long>((rows + rpb - 1) / rpb, 4096L);
     bn::bwd_stats_kernel_v<<<grid, 256, 0, stream>>>(
         (const bf16*)dy.data_ptr(), (const bf16*)x.data_ptr(),
         (const bf16*)y.data_ptr(), mean.data_ptr<float>(),
@@ -308,13 +383,25 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor x,
         sum_dyx.data_ptr<float>(), rows, c, relu ? 1 : 0);
   }
   auto dx = torch::empty_like(dy);
-  int grid1 = (int)std::min<long>((n + 255) / 256, 4096L);
-  bn::bwd_dx_kernel<<<grid1, 256, 0, stream>>>(
-      (const bf16*)dy.data_ptr(), (const bf16*)x.data_ptr(),
-      (const bf16*)y.data_ptr(), mean.data_ptr<float>(),
-      invstd.data_ptr<float>(), gamma.data_ptr<float>(),
-      sum_dy.data_ptr<float>(), sum_dyx.data_ptr<float>(),
-      (bf16*)dx.data_ptr(), n, rows, c, relu ? 1 : 0, training ? 1 : 0);
+  if (c % 8 == 0) {
+    long n8 = n / 8;
+    int grid1 = (int)std::min<long>((n8 + 255) / 256, 4096L);
+    bn::bwd_dx_kernel_v<<<grid1, 256, 0, stream>>>(
+        (const bf16*)dy.data_ptr(), (const bf16*)x.data_ptr(),
+        (const bf16*)y.data_ptr(), mean.data_ptr<float>(),
+        invstd.data_ptr<float>(), gamma.data_ptr<float>(),
+        sum_dy.data_ptr<float>(), sum_dyx.data_ptr<float>(),
+        (bf16*)dx.data_ptr(), n8, rows, c / 8, relu ? 1 : 0,
+        training ? 1 : 0);
+  } else {
+    int grid1 = (int)std::min<long>((n + 255) / 256, 4096L);
+    bn::bwd_dx_kernel<<<grid1, 256, 0, stream>>>(
+        (const bf16*)dy.data_ptr(), (const bf16*)x.data_ptr(),
+        (const bf16*)y.data_ptr(), mean.data_ptr<float>(),
+        invstd.data_ptr<float>(), gamma.data_ptr<float>(),
+        sum_dy.data_ptr<float>(), sum_dyx.data_ptr<float>(),
+        (bf16*)dx.data_ptr(), n, rows, c, relu ? 1 : 0, training ? 1 : 0);
+  }
   // dgamma = sum_dyx, dbeta = sum_dy (already per-channel fp32)
   return {dx, sum_dyx, sum_dy};
 }

@@ -108,25 +108,31 @@ __device__ __forceinline__ void row8(const bf16* g, long rows, long cols,
 }
 
 // ---- fwd & dgrad main loop ----------------------------------------------
-// 128x64 block tile, BK=64, 8 waves (4x2) of 32x32, 512 threads.
+// TBM x 64 block tile (TBM 128 for chip-filling shapes, 64 when the grid
+// would drop under ~1.5 blocks/CU), BK=64, TBM*4 threads = TBM/16 waves of
+// 32x32 wave-tiles. Software-pipelined register staging (guide G15/T14):
+// the NEXT tile's global loads are issued right after this tile's LDS
+// write, so HBM latency hides under the MFMA phase.
 
-template <int MODE, bool FAST>  // MODE: 0 = fwd, 1 = dgrad
-__global__ __launch_bounds__(512) void conv_mm_kernel(
+template <int MODE, bool FAST, int TBM>  // MODE: 0 = fwd, 1 = dgrad
+__global__ __launch_bounds__(TBM * 4) void conv_mm_kernel(
     const bf16* __restrict__ Asrc, const bf16* __restrict__ B,
     const float* __restrict__ bias, bf16* __restrict__ out, Geom g,
     long M, long N, long RED, int has_bias) {
-  __shared__ __bf16 sA[CBM * CLDK];
+  constexpr int THREADS = TBM * 4;
+  constexpr int BROWS = 512 / THREADS;  // B-staging rows per thread
+  __shared__ __bf16 sA[TBM * CLDK];
   __shared__ __bf16 sB[CBN * CLDK];
 
-  const long m0 = (long)blockIdx.x * CBM;
+  const long m0 = (long)blockIdx.x * TBM;
   const long n0 = (long)blockIdx.y * CBN;
   const int t = threadIdx.x;
   const int lane = t & 63;
   const int wave = t >> 6;
   const int wr = wave >> 1, wc = wave & 1;
-  const int lra = t >> 2;            // A staging row 0..127
+  const int lra = t >> 2;            // A staging row 0..TBM-1
   const int lka = (t & 3) * 8;       // A staging col {0,8,16,24}
-  const int lrb = t >> 3;            // B staging row 0..63
+  const int lrb = t >> 3;            // B staging row
   const int lkb = (t & 7) * 8;       // B staging col {0..56}
 
   // hoisted per-thread A-row pixel decomposition
@@ -152,10 +158,9 @@ __global__ __launch_bounds__(512) void conv_mm_kernel(
     cur_rs[p] = FAST ? red / INNER : 0;
   }
 
-  f32x4 acc[2][2] = {};
+  __bf16 ra[2][8], rb[BROWS][8];
 
-  for (long k0 = 0; k0 < RED; k0 += CBK) {
-    __bf16 ra[2][8], rb[8];
+  auto stage = [&](long k0) {
 #pragma unroll
     for (int p = 0; p < 2; ++p) {
       if (FAST) {
@@ -198,14 +203,28 @@ __global__ __launch_bounds__(512) void conv_mm_kernel(
         }
       }
     }
-    row8(B, N, RED, n0 + lrb, k0 + lkb, rb);
-    __syncthreads();
+#pragma unroll
+    for (int q = 0; q < BROWS; ++q) {
+      row8(B, N, RED, n0 + lrb + q * (THREADS / 8), k0 + lkb, rb[q]);
+    }
+  };
+
+  f32x4 acc[2][2] = {};
+  stage(0);
+
+  for (long k0 = 0; k0 < RED; k0 += CBK) {
+    __syncthreads();  // prior tile's fragment reads done; LDS reusable
     *reinterpret_cast<bf16x8*>(&sA[lra * CLDK + lka]) =
         *reinterpret_cast<bf16x8*>(ra[0]);
     *reinterpret_cast<bf16x8*>(&sA[lra * CLDK + lka + 32]) =
         *reinterpret_cast<bf16x8*>(ra[1]);
-    *reinterpret_cast<bf16x8*>(&sB[lrb * CLDK + lkb]) =
-        *reinterpret_cast<bf16x8*>(rb);
+#pragma unroll
+    for (int q = 0; q < BROWS; ++q) {
+      *reinterpret_cast<bf16x8*>(
+          &sB[(lrb + q * (THREADS / 8)) * CLDK + lkb]) =
+          *reinterpret_cast<bf16x8*>(rb[q]);
+    }
+    if (k0 + CBK < RED) stage(k0 + CBK);  // overlap loads with MFMA below
     __syncthreads();
 
     const int ml = lane & 15;
@@ -282,12 +301,17 @@ __device__ __forceinline__ bf16x8 tr16_frag(const __bf16* base, int np0,
   return r;
 }
 
+// image block stride: +16 elems (32 B) padding between the four 16-inner
+// blocks — without it the b128 staging writes of different kb-blocks land
+// on the same 8 banks (measured 4-way, SQ_LDS_BANK_CONFLICT ~= MFMA time).
+constexpr int WIMG = WBK * 16 + 16;
+
 __global__ __launch_bounds__(256) void conv_wgrad_fast_kernel(
     const bf16* __restrict__ x, const bf16* __restrict__ dy,
     float* __restrict__ dw, Geom g, long NP, long npslice) {
   // images: [inner16-block][np(64)][16] per operand (4 blocks of 64 k/rsc)
-  __shared__ __bf16 sA[4 * WBK * 16];
-  __shared__ __bf16 sB[4 * WBK * 16];
+  __shared__ __bf16 sA[4 * WIMG];
+  __shared__ __bf16 sB[4 * WIMG];
 
   const long k0c = (long)blockIdx.x * WBM;
   const long n0 = (long)blockIdx.y * WBN;
@@ -301,7 +325,7 @@ __global__ __launch_bounds__(256) void conv_wgrad_fast_kernel(
   const int wr = wave >> 1, wc = wave & 1;
   const int snp = t >> 3;          // 0..31 (stages rows snp, snp+32)
   const int scol = (t & 7) * 8;    // 0..56
-  const int simg = (scol >> 4) * (WBK * 16) + (scol & 15);
+  const int simg = (scol >> 4) * WIMG + (scol & 15);
 
   // fixed per-thread B column (r, s, c)
   int fb_r = 0, fb_s = 0, fb_c = 0;
@@ -382,11 +406,11 @@ __global__ __launch_bounds__(256) void conv_wgrad_fast_kernel(
 #pragma unroll
       for (int fi = 0; fi < 2; ++fi) {
         bf16x8 af =
-            tr16_frag(&sA[(wr * 2 + fi) * (WBK * 16)], ks * 32, lane);
+            tr16_frag(&sA[(wr * 2 + fi) * WIMG], ks * 32, lane);
 #pragma unroll
         for (int fj = 0; fj < 2; ++fj) {
           bf16x8 bfr =
-              tr16_frag(&sB[(wc * 2 + fj) * (WBK * 16)], ks * 32, lane);
+              tr16_frag(&sB[(wc * 2 + fj) * WIMG], ks * 32, lane);
           acc[fi][fj] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               af, bfr, acc[fi][fj], 0, 0, 0);
         }
@@ -616,10 +640,17 @@ torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor bias,
   long RED = (long)g.R * g.S * g.C;
   bool has_bias = bias.numel() > 0;
   bool fast = (g.C % 8 == 0);
-  dim3 grid(ceil_div(M, conv::CBM), ceil_div(g.K, conv::CBN));
-  auto* fn = fast ? conv::conv_mm_kernel<0, true>
-                  : conv::conv_mm_kernel<0, false>;
-  fn<<<grid, 512, 0, cur_stream()>>>(
+  // shape-adaptive tile: 128-row tiles starve the 256-CU chip on the deep
+  // stages (M drops to 4096); switch to 64-row tiles below ~384 blocks.
+  long blocks128 = (long)ceil_div(M, 128) * ceil_div(g.K, conv::CBN);
+  bool narrow = blocks128 < 384;
+  long tbm = narrow ? 64 : 128;
+  dim3 grid(ceil_div(M, tbm), ceil_div(g.K, conv::CBN));
+  auto* fn = fast ? (narrow ? conv::conv_mm_kernel<0, true, 64>
+                            : conv::conv_mm_kernel<0, true, 128>)
+                  : (narrow ? conv::conv_mm_kernel<0, false, 64>
+                            : conv::conv_mm_kernel<0, false, 128>);
+  fn<<<grid, (unsigned)(tbm * 4), 0, cur_stream()>>>(
       (const bf16*)x.data_ptr(), (const bf16*)w.data_ptr(),
       has_bias ? bias.data_ptr<float>() : nullptr, (bf16*)y.data_ptr(), g,
       M, g.K, RED, has_bias ? 1 : 0);
@@ -641,10 +672,15 @@ torch::Tensor conv2d_dgrad(torch::Tensor dy, torch::Tensor wt, long stride,
   long M = (long)g.N * g.H * g.W;
   long RED = (long)R * S * K;
   bool fast = (g.K % 8 == 0) && (g.stride <= 2);
-  dim3 grid(ceil_div(M, conv::CBM), ceil_div(C, conv::CBN));
-  auto* fn = fast ? conv::conv_mm_kernel<1, true>
-                  : conv::conv_mm_kernel<1, false>;
-  fn<<<grid, 512, 0, cur_stream()>>>(
+  long blocks128 = (long)ceil_div(M, 128) * ceil_div(C, conv::CBN);
+  bool narrow = blocks128 < 384;
+  long tbm = narrow ? 64 : 128;
+  dim3 grid(ceil_div(M, tbm), ceil_div(C, conv::CBN));
+  auto* fn = fast ? (narrow ? conv::conv_mm_kernel<1, true, 64>
+                            : conv::conv_mm_kernel<1, true, 128>)
+                  : (narrow ? conv::conv_mm_kernel<1, false, 64>
+                            : conv::conv_mm_kernel<1, false, 128>);
+  fn<<<grid, (unsigned)(tbm * 4), 0, cur_stream()>>>(
       (const bf16*)dy.data_ptr(), (const bf16*)wt.data_ptr(), nullptr,
       (bf16*)dx.data_ptr(), g, M, C, RED, 0);
   return dx;
