@@ -6,6 +6,8 @@
 
 #include "common.h"
 
+#include <unordered_map>
+
 namespace eg {
 
 namespace bn {
@@ -81,6 +83,8 @@ __global__ void finalize_kernel(float* __restrict__ s, float* __restrict__ sq,
   var = fmaxf(var, 0.f);
   mean[ch] = m;
   invstd[ch] = rsqrtf(var + eps);
+  s[ch] = 0.f;   // recycle the persistent workspace (no per-call zeros)
+  sq[ch] = 0.f;
   if (update_running) {
     float unbiased = rows > 1 ? var * rows / (rows - 1) : var;
     running_mean[ch] = (1.f - momentum) * running_mean[ch] + momentum * m;
@@ -318,7 +322,12 @@ std::vector<torch::Tensor> bn_fwd(torch::Tensor x, torch::Tensor gamma,
   auto invstd = torch::empty({c}, f32);
   auto stream = cur_stream();
   if (training) {
-    auto ssq = torch::zeros({2 * c}, f32);
+    // persistent per-C workspace, zeroed once and reset by finalize_kernel
+    static auto* ws_cache = new std::unordered_map<int, torch::Tensor>();
+    auto it = ws_cache->find(c);
+    if (it == ws_cache->end())
+      it = ws_cache->emplace(c, torch::zeros({2 * c}, f32)).first;
+    auto ssq = it->second;
     auto s = ssq.narrow(0, 0, c);
     auto sq = ssq.narrow(0, c, c);
     if (c % 8 == 0 && c <= bn::BN_MAXC) {

@@ -287,8 +287,8 @@ typedef bf16x4t __attribute__((address_space(3)))* lds_b64p;
 // coords (a = (l>>2)&3, b = l&3) must SUPPLY elem (np0 + 8g + a)*16 + 4b.
 __device__ __forceinline__ bf16x8 tr16_frag(const __bf16* base, int np0,
                                             int lane) {
-  const __bf16* p = base +
-      (np0 + ((lane >> 4) << 3) + ((lane >> 2) & 3)) * 16 + (lane & 3) * 4;
+  const int np = np0 + ((lane >> 4) << 3) + ((lane >> 2) & 3);
+  const __bf16* p = base + np * 16 + ((np >> 3) * 8) + (lane & 3) * 4;
   auto p3 = (lds_b64p)(__bf16 __attribute__((address_space(3)))*)p;
   bf16x4t v0 = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(p3);
   bf16x4t v1 = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(p3 + 16);
@@ -304,8 +304,12 @@ __device__ __forceinline__ bf16x8 tr16_frag(const __bf16* base, int np0,
 // image block stride: +16 elems (32 B) padding between the four 16-inner
 // blocks — without it the b128 staging writes of different kb-blocks land
 // on the same 8 banks (measured 4-way, SQ_LDS_BANK_CONFLICT ~= MFMA time).
-constexpr int WIMG = WBK * 16 + 16;
+// per-np-group padding: +8 elems (16 B) after every 8 np rows — without
+// it the two 16-lane tr-read groups of a wave land on identical banks.
+__device__ __forceinline__ int np_img(int np) { return np * 16 + (np >> 3) * 8; }
+constexpr int WIMG = WBK * 16 + (WBK / 8) * 8 + 16;
 
+template <bool ATOMIC>
 __global__ __launch_bounds__(256) void conv_wgrad_fast_kernel(
     const bf16* __restrict__ x, const bf16* __restrict__ dy,
     float* __restrict__ dw, Geom g, long NP, long npslice) {
@@ -394,9 +398,9 @@ __global__ __launch_bounds__(256) void conv_wgrad_fast_kernel(
 #pragma unroll
     for (int q = 0; q < 2; ++q) {
       const int npl = snp + 32 * q;
-      *reinterpret_cast<bf16x8*>(&sA[simg + npl * 16]) =
+      *reinterpret_cast<bf16x8*>(&sA[simg + np_img(npl)]) =
           *reinterpret_cast<bf16x8*>(ra[q]);
-      *reinterpret_cast<bf16x8*>(&sB[simg + npl * 16]) =
+      *reinterpret_cast<bf16x8*>(&sB[simg + np_img(npl)]) =
           *reinterpret_cast<bf16x8*>(rb[q]);
     }
     __syncthreads();
@@ -430,7 +434,11 @@ __global__ __launch_bounds__(256) void conv_wgrad_fast_kernel(
       for (int r = 0; r < 4; ++r) {
         long kk = k0c + wr * 32 + fi * 16 + cm + r;
         if (kk >= g.K) continue;
-        atomicAdd(&dw[kk * RED + nn], acc[fi][fj][r]);
+        if (ATOMIC) {
+          atomicAdd(&dw[kk * RED + nn], acc[fi][fj][r]);
+        } else {
+          dw[kk * RED + nn] = acc[fi][fj][r];
+        }
       }
     }
   }
@@ -695,9 +703,8 @@ torch::Tensor conv2d_wgrad(torch::Tensor x, torch::Tensor dy, long R, long S,
               "wgrad geometry mismatch");
   long NP = (long)g.N * g.Ho * g.Wo;
   long RED = (long)R * S * g.C;
-  auto dw = torch::zeros({(long)g.K, RED}, x.options().dtype(torch::kFloat32));
   // split the NP reduction across blocks for parallelism; fp32 atomics
-  long target_blocks = 1024;
+  long target_blocks = 2048;
   long tiles = (long)ceil_div(g.K, conv::WBM) * ceil_div(RED, conv::WBN);
   long zsplit =
       std::max(1L, std::min(512L, target_blocks / std::max(tiles, 1L)));
@@ -705,13 +712,23 @@ torch::Tensor conv2d_wgrad(torch::Tensor x, torch::Tensor dy, long R, long S,
   npslice = ((npslice + conv::WBK - 1) / conv::WBK) * conv::WBK;
   zsplit = (NP + npslice - 1) / npslice;
   bool fast = (g.K % 8 == 0) && (g.C % 8 == 0);
+  // single NP slice per tile -> plain stores, no zero-init needed
+  auto dw = zsplit == 1
+      ? torch::empty({(long)g.K, RED}, x.options().dtype(torch::kFloat32))
+      : torch::zeros({(long)g.K, RED}, x.options().dtype(torch::kFloat32));
   dim3 grid(ceil_div(g.K, conv::WBM), ceil_div(RED, conv::WBN),
             (unsigned)zsplit);
-  if (fast) {
-    conv::conv_wgrad_fast_kernel<<<grid, 256, 0, cur_stream()>>>(
+  if (fast && zsplit == 1) {
+    conv::conv_wgrad_fast_kernel<false><<<grid, 256, 0, cur_stream()>>>(
+        (const bf16*)x.data_ptr(), (const bf16*)dy.data_ptr(),
+        dw.data_ptr<float>(), g, NP, npslice);
+  } else if (fast) {
+    conv::conv_wgrad_fast_kernel<true><<<grid, 256, 0, cur_stream()>>>(
         (const bf16*)x.data_ptr(), (const bf16*)dy.data_ptr(),
         dw.data_ptr<float>(), g, NP, npslice);
   } else {
+    TORCH_CHECK(zsplit >= 1);
+    if (zsplit == 1) dw.zero_();
     conv::conv_wgrad_kernel<false><<<grid, 256, 0, cur_stream()>>>(
         (const bf16*)x.data_ptr(), (const bf16*)dy.data_ptr(),
         dw.data_ptr<float>(), g, NP, npslice);
