@@ -31,7 +31,7 @@ torch::Tensor krsc_to_oihw(torch::Tensor);
 torch::Tensor gemm_bias(torch::Tensor, torch::Tensor, torch::Tensor, bool);
 // conv.hip
 torch::Tensor conv2d_fwd(torch::Tensor, torch::Tensor, torch::Tensor, long,
-                         long);
+                         long, bool);
 torch::Tensor conv2d_dgrad(torch::Tensor, torch::Tensor, long, long, long,
                            long);
 torch::Tensor conv2d_wgrad(torch::Tensor, torch::Tensor, long, long, long,
@@ -39,7 +39,7 @@ torch::Tensor conv2d_wgrad(torch::Tensor, torch::Tensor, long, long, long,
 // bn.hip
 std::vector<torch::Tensor> bn_fwd(torch::Tensor, torch::Tensor, torch::Tensor,
                                   torch::Tensor, torch::Tensor, double,
-                                  double, bool, bool);
+                                  double, bool, bool, bool);
 std::vector<torch::Tensor> bn_bwd(torch::Tensor, torch::Tensor, torch::Tensor,
                                   torch::Tensor, torch::Tensor, torch::Tensor,
                                   bool, bool);
@@ -74,10 +74,15 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("krsc_to_crsk", &eg::krsc_to_crsk);
   m.def("krsc_to_oihw", &eg::krsc_to_oihw);
   m.def("gemm_bias", &eg::gemm_bias);
-  m.def("conv2d_fwd", &eg::conv2d_fwd);
+  m.def("conv2d_fwd", &eg::conv2d_fwd, py::arg("x"), py::arg("w"),
+        py::arg("bias"), py::arg("stride"), py::arg("pad"),
+        py::arg("collect_bn_stats") = false);
   m.def("conv2d_dgrad", &eg::conv2d_dgrad);
   m.def("conv2d_wgrad", &eg::conv2d_wgrad);
-  m.def("bn_fwd", &eg::bn_fwd);
+  m.def("bn_fwd", &eg::bn_fwd, py::arg("x"), py::arg("gamma"), py::arg("beta"),
+        py::arg("running_mean"), py::arg("running_var"), py::arg("momentum"),
+        py::arg("eps"), py::arg("training"), py::arg("relu"),
+        py::arg("have_stats") = false);
   m.def("bn_bwd", &eg::bn_bwd);
   m.def("maxpool2x2_fwd", &eg::maxpool2x2_fwd);
   m.def("maxpool2x2_bwd", &eg::maxpool2x2_bwd);

@@ -10,6 +10,22 @@
 
 namespace eg {
 
+// shared persistent per-channel-count BN stats workspace (zeroed once;
+// finalize_kernel resets it after each consume). Also written by the conv
+// forward epilogue when collect_bn_stats is set (conv.hip).
+static torch::Tensor& bn_ws_tensor(int c, const torch::TensorOptions& f32) {
+  static auto* cache = new std::unordered_map<int, torch::Tensor>();
+  auto it = cache->find(c);
+  if (it == cache->end())
+    it = cache->emplace(c, torch::zeros({2 * c}, f32)).first;
+  return it->second;
+}
+
+float* bn_stats_ws_ptr(int c, torch::TensorOptions opts) {
+  return bn_ws_tensor(c, opts.dtype(torch::kFloat32))
+      .data_ptr<float>();
+}
+
 namespace bn {
 
 // per-channel sum & sumsq of x (bf16 [rows, C]) -> fp32 [C] each.
@@ -313,7 +329,8 @@ std::vector<torch::Tensor> bn_fwd(torch::Tensor x, torch::Tensor gamma,
                                   torch::Tensor beta,
                                   torch::Tensor running_mean,
                                   torch::Tensor running_var, double momentum,
-                                  double eps, bool training, bool relu) {
+                                  double eps, bool training, bool relu,
+                                  bool have_stats) {
   CHECK_IN(x);
   int c = (int)x.size(-1);
   long rows = x.numel() / c;
@@ -322,15 +339,14 @@ std::vector<torch::Tensor> bn_fwd(torch::Tensor x, torch::Tensor gamma,
   auto invstd = torch::empty({c}, f32);
   auto stream = cur_stream();
   if (training) {
-    // persistent per-C workspace, zeroed once and reset by finalize_kernel
-    static auto* ws_cache = new std::unordered_map<int, torch::Tensor>();
-    auto it = ws_cache->find(c);
-    if (it == ws_cache->end())
-      it = ws_cache->emplace(c, torch::zeros({2 * c}, f32)).first;
-    auto ssq = it->second;
+    // persistent per-C workspace, zeroed once and reset by finalize_kernel;
+    // when have_stats the producing conv's epilogue already filled it.
+    auto ssq = bn_ws_tensor(c, f32);
     auto s = ssq.narrow(0, 0, c);
     auto sq = ssq.narrow(0, c, c);
-    if (c % 8 == 0 && c <= bn::BN_MAXC) {
+    if (have_stats) {
+      // sums already accumulated by the conv epilogue
+    } else if (c % 8 == 0 && c <= bn::BN_MAXC) {
       int rpb = 256 / (c / 8);
       int grid = (int)std::min<long>((rows + rpb - 1) / rpb, 1024L);
       bn::stats_kernel_v<<<grid, 256, 0, stream>>>(

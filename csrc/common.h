@@ -30,6 +30,9 @@ static inline hipStream_t cur_stream() {
 
 static inline int ceil_div(long a, long b) { return (int)((a + b - 1) / b); }
 
+// defined in bn.hip; used by conv.hip's fused-BN-stats epilogue
+float* bn_stats_ws_ptr(int c, torch::TensorOptions opts);
+
 // simple per-element hash RNG (wang hash), uniform in [0,1)
 __device__ __forceinline__ float hash_uniform(unsigned seed, unsigned idx) {
   unsigned h = seed ^ (idx * 0x9E3779B9u);

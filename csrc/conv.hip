@@ -114,15 +114,25 @@ __device__ __forceinline__ void row8(const bf16* g, long rows, long cols,
 // the NEXT tile's global loads are issued right after this tile's LDS
 // write, so HBM latency hides under the MFMA phase.
 
-template <int MODE, bool FAST, int TBM>  // MODE: 0 = fwd, 1 = dgrad
+template <int MODE, bool FAST, int TBM, bool STATS = false>
 __global__ __launch_bounds__(TBM * 4) void conv_mm_kernel(
     const bf16* __restrict__ Asrc, const bf16* __restrict__ B,
     const float* __restrict__ bias, bf16* __restrict__ out, Geom g,
-    long M, long N, long RED, int has_bias) {
+    long M, long N, long RED, int has_bias,
+    float* __restrict__ stats_ws = nullptr) {
   constexpr int THREADS = TBM * 4;
   constexpr int BROWS = 512 / THREADS;  // B-staging rows per thread
   __shared__ __bf16 sA[TBM * CLDK];
   __shared__ __bf16 sB[CBN * CLDK];
+  // per-block BN-stats accumulators (fused batch-norm statistics: the
+  // separate per-channel stats pass over y disappears when the consumer
+  // is a training-mode BatchNorm)
+  __shared__ float s_sum[STATS ? CBN : 1];
+  __shared__ float s_sq[STATS ? CBN : 1];
+  if (STATS && threadIdx.x < CBN) {
+    s_sum[threadIdx.x] = 0.f;
+    s_sq[threadIdx.x] = 0.f;
+  }
 
   const long m0 = (long)blockIdx.x * TBM;
   const long n0 = (long)blockIdx.y * CBN;
@@ -247,17 +257,37 @@ __global__ __launch_bounds__(TBM * 4) void conv_mm_kernel(
   const int cn = lane & 15;
   const int cm = (lane >> 4) * 4;
 #pragma unroll
-  for (int fi = 0; fi < 2; ++fi) {
+  for (int fj = 0; fj < 2; ++fj) {
+    long nn = n0 + wc * 32 + fj * 16 + cn;
+    if (nn >= N) continue;
+    float bv = has_bias ? bias[nn] : 0.f;
+    float psum = 0.f, psq = 0.f;
 #pragma unroll
-    for (int fj = 0; fj < 2; ++fj) {
-      long nn = n0 + wc * 32 + fj * 16 + cn;
-      if (nn >= N) continue;
-      float bv = has_bias ? bias[nn] : 0.f;
+    for (int fi = 0; fi < 2; ++fi) {
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         long mm = m0 + wr * 32 + fi * 16 + cm + r;
         if (mm >= M) continue;
-        out[mm * N + nn] = f2b(acc[fi][fj][r] + bv);
+        float v = acc[fi][fj][r] + bv;
+        out[mm * N + nn] = f2b(v);
+        if (STATS) {
+          psum += v;
+          psq += v * v;
+        }
+      }
+    }
+    if (STATS) {
+      int nl = wc * 32 + fj * 16 + cn;
+      atomicAdd(&s_sum[nl], psum);
+      atomicAdd(&s_sq[nl], psq);
+    }
+  }
+  if (STATS) {
+    __syncthreads();
+    if (threadIdx.x < CBN && n0 + threadIdx.x < N) {
+      if (s_sum[threadIdx.x] != 0.f || s_sq[threadIdx.x] != 0.f) {
+        atomicAdd(&stats_ws[n0 + threadIdx.x], s_sum[threadIdx.x]);
+        atomicAdd(&stats_ws[N + n0 + threadIdx.x], s_sq[threadIdx.x]);
       }
     }
   }
@@ -635,8 +665,10 @@ static conv::Geom make_geom(const torch::Tensor& x, int K, int R, int S,
 }
 
 // x [N,H,W,C] bf16; w [K,R,S,C] bf16; bias fp32[K] or empty -> y [N,Ho,Wo,K]
+// collect_bn_stats: also accumulate per-out-channel sum/sumsq of y into the
+// shared BN workspace (bn.hip bn_stats_ws), consumed by bn_fwd(have_stats).
 torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor bias,
-                         long stride, long pad) {
+                         long stride, long pad, bool collect_bn_stats) {
   CHECK_IN(x); CHECK_IN(w);
   TORCH_CHECK(x.scalar_type() == torch::kBFloat16 &&
               w.scalar_type() == torch::kBFloat16);
@@ -654,14 +686,27 @@ torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor bias,
   bool narrow = blocks128 < 384;
   long tbm = narrow ? 64 : 128;
   dim3 grid(ceil_div(M, tbm), ceil_div(g.K, conv::CBN));
-  auto* fn = fast ? (narrow ? conv::conv_mm_kernel<0, true, 64>
-                            : conv::conv_mm_kernel<0, true, 128>)
-                  : (narrow ? conv::conv_mm_kernel<0, false, 64>
-                            : conv::conv_mm_kernel<0, false, 128>);
-  fn<<<grid, (unsigned)(tbm * 4), 0, cur_stream()>>>(
-      (const bf16*)x.data_ptr(), (const bf16*)w.data_ptr(),
-      has_bias ? bias.data_ptr<float>() : nullptr, (bf16*)y.data_ptr(), g,
-      M, g.K, RED, has_bias ? 1 : 0);
+  float* ws = nullptr;
+  if (collect_bn_stats) ws = bn_stats_ws_ptr(g.K, x.options());
+  if (ws != nullptr) {
+    auto* fn = fast ? (narrow ? conv::conv_mm_kernel<0, true, 64, true>
+                              : conv::conv_mm_kernel<0, true, 128, true>)
+                    : (narrow ? conv::conv_mm_kernel<0, false, 64, true>
+                              : conv::conv_mm_kernel<0, false, 128, true>);
+    fn<<<grid, (unsigned)(tbm * 4), 0, cur_stream()>>>(
+        (const bf16*)x.data_ptr(), (const bf16*)w.data_ptr(),
+        has_bias ? bias.data_ptr<float>() : nullptr, (bf16*)y.data_ptr(), g,
+        M, g.K, RED, has_bias ? 1 : 0, ws);
+  } else {
+    auto* fn = fast ? (narrow ? conv::conv_mm_kernel<0, true, 64>
+                              : conv::conv_mm_kernel<0, true, 128>)
+                    : (narrow ? conv::conv_mm_kernel<0, false, 64>
+                              : conv::conv_mm_kernel<0, false, 128>);
+    fn<<<grid, (unsigned)(tbm * 4), 0, cur_stream()>>>(
+        (const bf16*)x.data_ptr(), (const bf16*)w.data_ptr(),
+        has_bias ? bias.data_ptr<float>() : nullptr, (bf16*)y.data_ptr(), g,
+        M, g.K, RED, has_bias ? 1 : 0, nullptr);
+  }
   return y;
 }
 
@@ -690,7 +735,7 @@ torch::Tensor conv2d_dgrad(torch::Tensor dy, torch::Tensor wt, long stride,
                             : conv::conv_mm_kernel<1, false, 128>);
   fn<<<grid, (unsigned)(tbm * 4), 0, cur_stream()>>>(
       (const bf16*)dy.data_ptr(), (const bf16*)wt.data_ptr(), nullptr,
-      (bf16*)dx.data_ptr(), g, M, C, RED, 0);
+      (bf16*)dx.data_ptr(), g, M, C, RED, 0, nullptr);
   return dx;
 }
 

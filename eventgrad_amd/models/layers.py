@@ -38,8 +38,9 @@ class Conv2d(nn.Module):
             bound = 1.0 / math.sqrt(fan_in)
             nn.init.uniform_(self.bias, -bound, bound)
 
-    def forward(self, x):
-        return O.conv2d(x, self.weight, self.bias, self.stride, self.padding)
+    def forward(self, x, bn_stats: bool = False):
+        return O.conv2d(x, self.weight, self.bias, self.stride, self.padding,
+                        bn_stats=bn_stats and x.is_cuda and self.training)
 
     def extra_repr(self):
         return (f"{self.in_ch}, {self.out_ch}, k={self.kernel_size}, "
@@ -67,12 +68,13 @@ class BatchNorm2d(nn.Module):
         super()._load_from_state_dict(*args, **kwargs)
         self._nbt = int(self.num_batches_tracked.item())
 
-    def forward(self, x, fuse_relu: bool = False):
+    def forward(self, x, fuse_relu: bool = False, stats_ready: bool = False):
         if self.training:
             self._nbt += 1
         return O.batch_norm(x, self.weight, self.bias, self.running_mean,
                             self.running_var, self.training, self.momentum,
-                            self.eps, fuse_relu)
+                            self.eps, fuse_relu,
+                            stats_ready and x.is_cuda and self.training)
 
 
 class Linear(nn.Module):

@@ -39,7 +39,7 @@ class Downsampler(nn.Module):
         self.bn = BatchNorm2d(out_ch)
 
     def forward(self, x):
-        return self.bn(self.conv(x))
+        return self.bn(self.conv(x, bn_stats=True), stats_ready=True)
 
 
 class BasicBlock(nn.Module):
@@ -55,8 +55,9 @@ class BasicBlock(nn.Module):
                             if downsample else None)
 
     def forward(self, x):
-        out = self.bn1(self.conv1(x), fuse_relu=True)
-        out = self.bn2(self.conv2(out))
+        out = self.bn1(self.conv1(x, bn_stats=True), fuse_relu=True,
+                       stats_ready=True)
+        out = self.bn2(self.conv2(out, bn_stats=True), stats_ready=True)
         residual = self.downsampler(x) if self.downsampler is not None else x
         return O.add_relu(out, residual)
 
@@ -76,9 +77,11 @@ class BottleNeck(nn.Module):
                             if downsample else None)
 
     def forward(self, x):
-        out = self.bn1(self.conv1(x), fuse_relu=True)
-        out = self.bn2(self.conv2(out), fuse_relu=True)
-        out = self.bn3(self.conv3(out))
+        out = self.bn1(self.conv1(x, bn_stats=True), fuse_relu=True,
+                       stats_ready=True)
+        out = self.bn2(self.conv2(out, bn_stats=True), fuse_relu=True,
+                       stats_ready=True)
+        out = self.bn3(self.conv3(out, bn_stats=True), stats_ready=True)
         residual = self.downsampler(x) if self.downsampler is not None else x
         return O.add_relu(out, residual)
 
@@ -111,7 +114,8 @@ class ResNet(nn.Module):
 
     def forward(self, x):
         x = O.to_compute(x)
-        out = self.bn(self.conv(x), fuse_relu=True)
+        out = self.bn(self.conv(x, bn_stats=True), fuse_relu=True,
+                      stats_ready=True)
         out = self.layer1(out)
         out = self.layer2(out)
         out = self.layer3(out)

@@ -63,11 +63,11 @@ def _w_krsc(w: torch.Tensor) -> torch.Tensor:
 
 class _ConvNHWC(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, x, w, bias, stride, padding):
+    def forward(ctx, x, w, bias, stride, padding, bn_stats):
         core = native()
         wk = _w_krsc(w)
         b = bias.detach() if bias is not None else _empty_f32(x.device)
-        y = core.conv2d_fwd(x, wk, b, stride, padding)
+        y = core.conv2d_fwd(x, wk, b, stride, padding, bn_stats)
         ctx.save_for_backward(x, wk)
         ctx.stride, ctx.padding = stride, padding
         ctx.has_bias = bias is not None
@@ -91,12 +91,16 @@ class _ConvNHWC(torch.autograd.Function):
             dw = core.krsc_to_oihw(dw_krsc)  # KRSC fp32 -> OIHW fp32
         if ctx.has_bias and ctx.needs_input_grad[2]:
             db = core.channel_sum(dy)
-        return dx, dw, db, None, None
+        return dx, dw, db, None, None, None
 
 
-def conv2d(x, w, bias=None, stride=1, padding=0):
+def conv2d(x, w, bias=None, stride=1, padding=0, bn_stats=False):
+    """bn_stats: fuse the following training-mode BatchNorm's batch-stats
+    accumulation into this conv's epilogue (pair with
+    batch_norm(..., stats_ready=True))."""
     if x.is_cuda:
-        return _ConvNHWC.apply(x, w, bias, int(stride), int(padding))
+        return _ConvNHWC.apply(x, w, bias, int(stride), int(padding),
+                               bool(bn_stats))
     return F.conv2d(x, w, bias, stride=stride, padding=padding)
 
 
@@ -107,11 +111,11 @@ def conv2d(x, w, bias=None, stride=1, padding=0):
 class _BatchNormNHWC(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, gamma, beta, running_mean, running_var, training,
-                momentum, eps, fuse_relu):
+                momentum, eps, fuse_relu, stats_ready):
         core = native()
         y, save_mean, save_invstd = core.bn_fwd(
             x, gamma.detach(), beta.detach(), running_mean, running_var,
-            momentum, eps, training, fuse_relu)
+            momentum, eps, training, fuse_relu, stats_ready)
         ctx.save_for_backward(x, gamma, save_mean, save_invstd, y)
         ctx.fuse_relu = fuse_relu
         ctx.training = training
@@ -124,14 +128,16 @@ class _BatchNormNHWC(torch.autograd.Function):
         dx, dgamma, dbeta = core.bn_bwd(dy.contiguous(), x, save_mean,
                                         save_invstd, gamma.detach(), y,
                                         ctx.fuse_relu, ctx.training)
-        return dx, dgamma, dbeta, None, None, None, None, None, None
+        return (dx, dgamma, dbeta, None, None, None, None, None, None,
+                None)
 
 
 def batch_norm(x, gamma, beta, running_mean, running_var, training,
-               momentum=0.1, eps=1e-5, fuse_relu=False):
+               momentum=0.1, eps=1e-5, fuse_relu=False, stats_ready=False):
     if x.is_cuda:
         return _BatchNormNHWC.apply(x, gamma, beta, running_mean, running_var,
-                                    training, momentum, eps, fuse_relu)
+                                    training, momentum, eps, fuse_relu,
+                                    stats_ready and training)
     y = F.batch_norm(x, running_mean, running_var, gamma, beta, training,
                      momentum, eps)
     return F.relu(y) if fuse_relu else y

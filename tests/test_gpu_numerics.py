@@ -339,3 +339,28 @@ def test_topk_absdiff():
     untouched = torch.ones(n, dtype=torch.bool, device=DEV)
     untouched[idx.long()] = False
     assert torch.equal(prev[untouched], prev_copy[untouched])
+
+
+def test_conv_fused_bn_stats_matches_separate():
+    """conv(bn_stats=True) + bn_fwd(have_stats) == conv + separate stats."""
+    torch.manual_seed(21)
+    from eventgrad_amd.ops import functional as O
+    N, H, W, C, K = 4, 16, 16, 32, 64
+    x = torch.randn(N, H, W, C, device=DEV).to(torch.bfloat16)
+    w = torch.randn(K, C, 3, 3, device=DEV) * 0.05
+    gamma = torch.rand(K, device=DEV)
+    beta = torch.randn(K, device=DEV)
+
+    def run(fused):
+        rm = torch.zeros(K, device=DEV)
+        rv = torch.ones(K, device=DEV)
+        y = O.conv2d(x.clone(), w, None, 1, 1, bn_stats=fused)
+        out = O.batch_norm(y, gamma, beta, rm, rv, True, 0.1, 1e-5,
+                           fuse_relu=True, stats_ready=fused)
+        return out.float(), rm.clone(), rv.clone()
+
+    o1, rm1, rv1 = run(False)
+    o2, rm2, rv2 = run(True)
+    assert rel_err(o2, o1) < 1e-2
+    assert rel_err(rm2, rm1) < 1e-3
+    assert rel_err(rv2, rv1) < 1e-3
