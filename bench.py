@@ -41,6 +41,8 @@ def main() -> int:
     ap.add_argument("--mode", default="event",
                     choices=["event", "decent", "cent", "spevent", "serial"])
     ap.add_argument("--model", default="resnet18q")
+    ap.add_argument("--no-graph", action="store_true",
+                    help="disable hipGraph capture of fwd+bwd")
     args = ap.parse_args()
 
     cfg = preset("dcifar10-event")
@@ -68,16 +70,24 @@ def main() -> int:
           for _ in range(n_resident)]
 
     pass_num = 0
+    graph = None
+    if not args.no_graph:
+        from eventgrad_amd.train.graphstep import FwdBwdGraph, can_graph
+        if can_graph(model, device):
+            graph = FwdBwdGraph(model, space, tuple(xs[0].shape), device)
 
     def step():
         nonlocal pass_num
         pass_num += 1
         x, y = xs[pass_num % n_resident], ys[pass_num % n_resident]
         engine.begin_pass(pass_num)
-        space.zero_grad()
-        logits = model(x)
-        loss = O.nll_of_logits(logits, y)
-        loss.backward()
+        if graph is not None:
+            logits, loss = graph.step(x, y)
+        else:
+            space.zero_grad()
+            logits = model(x)
+            loss = O.nll_of_logits(logits, y)
+            loss.backward()
         engine.after_backward()
         engine.step()
         return loss
@@ -126,6 +136,7 @@ def main() -> int:
                 "parallelism": f"eventgrad-ring dp{world}",
                 "optimizer": "sgd lr=1e-2 momentum=0.9",
                 "trigger": "adaptive horizon=1.01 warmup=30",
+                "hip_graph": graph is not None,
                 "final_loss": round(float(loss.item()), 4),
             },
         }

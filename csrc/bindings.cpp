@@ -50,7 +50,7 @@ torch::Tensor avgpool_fwd(torch::Tensor, long);
 torch::Tensor avgpool_bwd(torch::Tensor, long, long, long);
 // loss.hip
 std::vector<torch::Tensor> logsoftmax_nll_fwd(torch::Tensor, torch::Tensor);
-torch::Tensor logsoftmax_nll_bwd(torch::Tensor, torch::Tensor, double);
+torch::Tensor logsoftmax_nll_bwd(torch::Tensor, torch::Tensor, torch::Tensor);
 // topk.hip
 std::vector<torch::Tensor> topk_absdiff(torch::Tensor, torch::Tensor, long);
 void scatter_update(torch::Tensor, torch::Tensor, torch::Tensor);

@@ -26,9 +26,11 @@ __global__ void lsm_nll_fwd_kernel(const bf16* __restrict__ logits,
 __global__ void lsm_nll_bwd_kernel(const float* __restrict__ logp,
                                    const long* __restrict__ target,
                                    float* __restrict__ dlogits, int B, int C,
-                                   float gscale) {
+                                   const float* __restrict__ dloss) {
   int i = blockIdx.x * blockDim.x + threadIdx.x;
   if (i >= B * C) return;
+  // dloss read on device (graph-capture safe: no host sync)
+  float gscale = dloss[0] / (float)B;
   int row = i / C, j = i % C;
   float soft = __expf(logp[i]);
   float g = (soft - (j == target[row] ? 1.f : 0.f)) * gscale;
@@ -50,14 +52,13 @@ std::vector<torch::Tensor> logsoftmax_nll_fwd(torch::Tensor logits,
 }
 
 torch::Tensor logsoftmax_nll_bwd(torch::Tensor logp, torch::Tensor target,
-                                 double dloss) {
-  CHECK_IN(logp); CHECK_IN(target);
+                                 torch::Tensor dloss) {
+  CHECK_IN(logp); CHECK_IN(target); CHECK_DEV(dloss);
   int B = (int)logp.size(0), C = (int)logp.size(1);
   auto d = torch::empty_like(logp);
-  float gscale = (float)(dloss / B);
   lsm_nll_bwd_kernel<<<ceil_div((long)B * C, 256), 256, 0, cur_stream()>>>(
       logp.data_ptr<float>(), target.data_ptr<long>(), d.data_ptr<float>(),
-      B, C, gscale);
+      B, C, dloss.data_ptr<float>());
   return d;
 }
 

@@ -97,6 +97,8 @@ class RunConfig:
     # device/dtype
     device: str = "auto"          # auto | cpu | cuda
     compute_dtype: str = "bf16"   # bf16 | fp32 (GPU compute dtype; params fp32)
+    hip_graph: bool = True        # capture fwd+bwd into a hipGraph on GPU
+    #                               (auto-skipped for models with dropout)
     # evaluation
     eval_at_end: bool = True
     final_consensus: bool = True  # closing param AllReduce (event.cpp:517-525)

@@ -23,6 +23,8 @@ from .layers import Conv2d, Linear
 
 
 class _MnistCNN(nn.Module):
+    uses_dropout = True  # per-step host RNG -> not hipGraph-capturable
+
     def __init__(self, k: int, fc_in: int, fc_mid: int):
         super().__init__()
         self.conv1 = Conv2d(1, 10, k)
@@ -57,6 +59,8 @@ class CNN1(_MnistCNN):
 
 
 class LeNet5(nn.Module):
+    uses_dropout = True
+
     def __init__(self):
         super().__init__()
         self.conv1 = Conv2d(3, 6, 5)

@@ -324,7 +324,9 @@ class _LogSoftmaxNLL(torch.autograd.Function):
     @staticmethod
     def backward(ctx, dloss):
         logp, target = ctx.saved_tensors
-        d = native().logsoftmax_nll_bwd(logp, target, float(dloss.item()))
+        # device-scalar dloss: no host sync (hipGraph-capture safe)
+        d = native().logsoftmax_nll_bwd(
+            logp, target, dloss.to(torch.float32).reshape(1).contiguous())
         return d.to(ctx.out_dtype), None
 
 

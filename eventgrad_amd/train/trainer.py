@@ -21,6 +21,7 @@ from ..models import build_model
 from ..ops import functional as O
 from ..parallel import FlatParamSpace, build_engine, init_distributed
 from .checkpoint import load_checkpoint, save_checkpoint
+from .graphstep import FwdBwdGraph, can_graph
 from .metrics import RunMetrics
 from .trace import Tracer
 
@@ -67,6 +68,8 @@ class Trainer:
     def train(self) -> RunMetrics:
         cfg = self.cfg
         self.model.train()
+        graph = None
+        use_graph = cfg.hip_graph and can_graph(self.model, self.device)
         t0 = time.perf_counter()
         for epoch in range(self.start_epoch, cfg.epochs + 1):
             correct = seen = 0
@@ -76,10 +79,16 @@ class Trainer:
                 y = y.to(self.device, non_blocking=True)
                 self.pass_num += 1
                 self.engine.begin_pass(self.pass_num)
-                self.space.zero_grad()
-                logits = self.model(x)
-                loss = O.nll_of_logits(logits, y)
-                loss.backward()
+                if use_graph:
+                    if graph is None:
+                        graph = FwdBwdGraph(self.model, self.space,
+                                            tuple(x.shape), self.device)
+                    logits, loss = graph.step(x, y)
+                else:
+                    self.space.zero_grad()
+                    logits = self.model(x)
+                    loss = O.nll_of_logits(logits, y)
+                    loss.backward()
                 self.engine.after_backward()
                 self.engine.step()
                 correct += O.accuracy_count(logits.detach(), y)
