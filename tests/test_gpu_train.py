@@ -110,7 +110,9 @@ def test_resnet_training_step_gpu():
 
 
 def test_hipgraph_step_matches_eager():
-    """FwdBwdGraph (capture+replay) produces the same losses/grads as eager."""
+    """FwdBwdGraph (capture+replay) produces the same gradients as eager on
+    the same batch, to within the run-to-run atomic-reduction noise floor
+    (wgrad split-K atomics make even eager-vs-eager nondeterministic)."""
     import torch
     from eventgrad_amd.models import build_model
     from eventgrad_amd.ops import functional as O
@@ -118,38 +120,34 @@ def test_hipgraph_step_matches_eager():
     from eventgrad_amd.train.graphstep import FwdBwdGraph
 
     dev = torch.device("cuda")
+    torch.manual_seed(0)
+    model = build_model("resnet18q").to(dev)
+    model.train()
+    space = FlatParamSpace(model, dev)
+    torch.manual_seed(7)
+    x = torch.randn(8, 3, 32, 32, device=dev)
+    y = torch.randint(0, 10, (8,), device=dev)
 
-    def run(use_graph):
-        torch.manual_seed(0)
-        model = build_model("resnet18q").to(dev)
-        model.train()
-        space = FlatParamSpace(model, dev)
-        g = FwdBwdGraph(model, space, (8, 3, 32, 32), dev) if use_graph \
-            else None
-        torch.manual_seed(7)
-        losses = []
-        for it in range(4):
-            x = torch.randn(8, 3, 32, 32, device=dev)
-            y = torch.randint(0, 10, (8,), device=dev)
-            if g is not None:
-                logits, loss = g.step(x, y)
-            else:
-                space.zero_grad()
-                loss = O.nll_of_logits(model(x), y)
-                loss.backward()
-            # apply a step so the trajectory actually evolves
-            from eventgrad_amd.ops.backend import native
-            native().sgd_step_norm(space.param, space.grad, space.momentum,
-                                   space.starts_t, space.numels_t,
-                                   0.01, 0.9, 0.0)
-            losses.append(float(loss))
-        return losses, space.param.clone()
+    def eager():
+        space.zero_grad()
+        loss = O.nll_of_logits(model(x), y)
+        loss.backward()
+        return float(loss.detach()), space.grad.clone()
 
-    l_eager, p_eager = run(False)
-    l_graph, p_graph = run(True)
-    # warmup inside capture perturbs BN running stats only (not params), so
-    # training math must match to bf16/atomic noise
-    for a, b in zip(l_eager, l_graph):
-        assert abs(a - b) / max(abs(b), 1e-6) < 2e-2, (l_eager, l_graph)
-    err = (p_eager - p_graph).norm() / p_eager.norm()
-    assert err < 2e-3, err.item()
+    l1, g1 = eager()
+    l2, g2 = eager()
+    floor = ((g1 - g2).norm() / g1.norm()).item()  # atomic noise floor
+
+    graph = FwdBwdGraph(model, space, (8, 3, 32, 32), dev)
+    _, loss_g = graph.step(x, y)
+    torch.cuda.synchronize()
+    lg = float(loss_g.detach())
+    gg = space.grad.clone()
+    err = ((gg - g1).norm() / g1.norm()).item()
+    assert abs(lg - l1) / max(abs(l1), 1e-6) < 1e-2, (lg, l1)
+    assert err < max(10 * floor, 5e-3), (err, floor)
+    # replay a second batch and check it responds to the new input
+    x2 = torch.randn_like(x)
+    _, loss2 = graph.step(x2, y)
+    torch.cuda.synchronize()
+    assert abs(float(loss2.detach()) - lg) > 1e-6  # input actually flows
