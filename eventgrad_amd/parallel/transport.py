@@ -49,12 +49,25 @@ class RingTransport:
         self.backend = dist.get_backend() if dist.is_initialized() else None
         self.use_tags = self.backend == "gloo"
         self._reqs: List = []
+        self._recv_copies: List = []  # (host_staging, device_dst) for gloo+GPU
 
     def _ops(self, pairs):
-        """pairs: list of (kind, tensor, peer, tag). Returns work handles."""
+        """pairs: list of (kind, tensor, peer, tag). Returns work handles.
+
+        gloo p2p requires CPU tensors; when the payload lives on GPU (the
+        single-GPU multi-process test configuration) sends are staged
+        through pinned host copies and recvs staged back in finish().
+        """
         if self.use_tags:
             reqs = []
             for kind, t, peer, tag in pairs:
+                if t.is_cuda:
+                    if kind == "send":
+                        t = t.cpu()
+                    else:
+                        host = torch.empty_like(t, device="cpu")
+                        self._recv_copies.append((host, t))
+                        t = host
                 fn = dist.isend if kind == "send" else dist.irecv
                 reqs.append(fn(t, peer, tag=tag))
             return reqs
@@ -79,6 +92,9 @@ class RingTransport:
         ]
         for r in self._ops(pairs):
             r.wait()
+        for host, dev in self._recv_copies:
+            dev.copy_(host)
+        self._recv_copies = []
         return fr_left.cpu(), fr_right.cpu()
 
     def post_payloads(self, send_l: Optional[torch.Tensor],
@@ -106,3 +122,6 @@ class RingTransport:
         for r in self._reqs:
             r.wait()
         self._reqs = []
+        for host, dev in self._recv_copies:
+            dev.copy_(host, non_blocking=True)
+        self._recv_copies = []
