@@ -30,6 +30,8 @@ def build_parser() -> argparse.ArgumentParser:
     p.add_argument("--epochs", type=int, default=None)
     p.add_argument("--batch-size", type=int, default=None)
     p.add_argument("--global-batch", type=int, default=None)
+    p.add_argument("--synthetic-samples", type=int, default=None,
+                   help="synthetic dataset train size (no-network runs)")
     p.add_argument("--lr", type=float, default=None)
     p.add_argument("--momentum", type=float, default=None)
     p.add_argument("--seed", type=int, default=None)
@@ -83,6 +85,9 @@ def config_from_args(args) -> RunConfig:
         cfg.data.batch_size = args.batch_size
     if args.global_batch is not None:
         cfg.data.global_batch = args.global_batch
+    if args.synthetic_samples is not None:
+        cfg.data.synthetic_train_samples = args.synthetic_samples
+        cfg.data.synthetic_test_samples = max(64, args.synthetic_samples // 8)
     if args.lr is not None:
         cfg.optim.lr = args.lr
     if args.momentum is not None:
