@@ -277,9 +277,17 @@ __global__ __launch_bounds__(TBM * 4) void conv_mm_kernel(
       }
     }
     if (STATS) {
-      int nl = wc * 32 + fj * 16 + cn;
-      atomicAdd(&s_sum[nl], psum);
-      atomicAdd(&s_sq[nl], psq);
+      // lanes l, l+16, l+32, l+48 hold the same output channel: reduce
+      // across the wave first so only 16 lanes touch the LDS accumulators
+      psum += __shfl_down(psum, 32);
+      psq += __shfl_down(psq, 32);
+      psum += __shfl_down(psum, 16);
+      psq += __shfl_down(psq, 16);
+      if ((lane >> 4) == 0) {
+        int nl = wc * 32 + fj * 16 + cn;
+        atomicAdd(&s_sum[nl], psum);
+        atomicAdd(&s_sq[nl], psq);
+      }
     }
   }
   if (STATS) {
@@ -687,7 +695,13 @@ torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor bias,
   long tbm = narrow ? 64 : 128;
   dim3 grid(ceil_div(M, tbm), ceil_div(g.K, conv::CBN));
   float* ws = nullptr;
-  if (collect_bn_stats) ws = bn_stats_ws_ptr(g.K, x.options());
+  if (collect_bn_stats) {
+    // the stats epilogue's wave-shuffle reduction assumes no masked
+    // output channels in any wave (true for every BN'd conv: K % 64 == 0)
+    TORCH_CHECK(g.K % 64 == 0,
+                "collect_bn_stats requires out-channels % 64 == 0");
+    ws = bn_stats_ws_ptr(g.K, x.options());
+  }
   if (ws != nullptr) {
     auto* fn = fast ? (narrow ? conv::conv_mm_kernel<0, true, 64, true>
                               : conv::conv_mm_kernel<0, true, 128, true>)
