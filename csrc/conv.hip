@@ -347,20 +347,16 @@ __device__ __forceinline__ bf16x8 tr16_frag(const __bf16* base, int np0,
 __device__ __forceinline__ int np_img(int np) { return np * 16 + (np >> 3) * 8; }
 constexpr int WIMG = WBK * 16 + (WBK / 8) * 8 + 16;
 
-template <bool ATOMIC, int TBN = 64>
+template <bool ATOMIC>
 __global__ __launch_bounds__(256) void conv_wgrad_fast_kernel(
     const bf16* __restrict__ x, const bf16* __restrict__ dy,
     float* __restrict__ dw, Geom g, long NP, long npslice) {
-  // images: [inner16-block][np(64)][16] per operand; A fixed 64 k-channels
-  // (4 blocks), B TBN rsc columns (TBN/16 blocks). TBN=128 halves the
-  // dy/x re-staging of the large-RED shapes.
-  constexpr int NBB = TBN / 16;      // B image kb-blocks
-  constexpr int FJ = TBN / 32;       // b-fragments per wave
+  // images: [inner16-block][np(64)][16] per operand (4 blocks of 64 k/rsc)
   __shared__ __bf16 sA[4 * WIMG];
-  __shared__ __bf16 sB[NBB * WIMG];
+  __shared__ __bf16 sB[4 * WIMG];
 
   const long k0c = (long)blockIdx.x * WBM;
-  const long n0 = (long)blockIdx.y * TBN;
+  const long n0 = (long)blockIdx.y * WBN;
   const long np0 = (long)blockIdx.z * npslice;
   const long np1 = min(np0 + npslice, NP);
   const long RED = (long)g.R * g.S * g.C;
@@ -369,62 +365,38 @@ __global__ __launch_bounds__(256) void conv_wgrad_fast_kernel(
   const int lane = t & 63;
   const int wave = t >> 6;
   const int wr = wave >> 1, wc = wave & 1;
-  // A staging: 32 np-rows x 8 col-chunks; rows {snp, snp+32}
-  const int snp = t >> 3;
-  const int scol = (t & 7) * 8;
+  const int snp = t >> 3;          // 0..31 (stages rows snp, snp+32)
+  const int scol = (t & 7) * 8;    // 0..56
   const int simg = (scol >> 4) * WIMG + (scol & 15);
-  // B staging: TBN cols: SBC threads per row; rows stride 256/SBC
-  constexpr int SBC = TBN / 8;       // threads per np row (8 or 16)
-  constexpr int QB = 64 / (256 / SBC);  // staging passes (2 or 4)
-  const int snpb = t / SBC;
-  const int scolb = (t % SBC) * 8;
-  const int simgb = (scolb >> 4) * WIMG + (scolb & 15);
 
-  // fixed per-thread B column (r, s, c) at scolb
+  // fixed per-thread B column (r, s, c)
   int fb_r = 0, fb_s = 0, fb_c = 0;
-  const bool bcol_ok = (n0 + scolb) < RED;
+  const bool bcol_ok = (n0 + scol) < RED;
   if (bcol_ok) {
-    int rsc = (int)(n0 + scolb);
+    int rsc = (int)(n0 + scol);
     fb_c = rsc % g.C;
     int rs = rsc / g.C;
     fb_r = rs / g.S;
     fb_s = rs % g.S;
   }
-  const bool b_vec_ok = bcol_ok && (fb_c + 8 <= g.C);
   const bool a_ok = k0c + scol + 8 <= g.K;
 
-  // incremental pixel cursors: A rows {snp, snp+32}; B rows {snpb + q*64/QB}
-  int pnA[2], phoA[2], pwoA[2];
-  int pnB[QB], phoB[QB], pwoB[QB];
-  auto init_cursor = [&](long m, int& n, int& ho, int& wo) {
+  // incremental pixel cursors for rows snp and snp+32
+  int pn[2], pho[2], pwo[2];
+#pragma unroll
+  for (int q = 0; q < 2; ++q) {
+    long m = np0 + snp + 32 * q;
     long mm = m < NP ? m : 0;
-    n = (int)(mm / ((long)g.Ho * g.Wo));
+    pn[q] = (int)(mm / ((long)g.Ho * g.Wo));
     int rem = (int)(mm % ((long)g.Ho * g.Wo));
-    ho = rem / g.Wo;
-    wo = rem % g.Wo;
-  };
-  auto adv_cursor = [&](int& ho, int& wo, int& n) {
-    wo += WBK;
-    while (wo >= g.Wo) {
-      wo -= g.Wo;
-      ho += 1;
-    }
-    while (ho >= g.Ho) {
-      ho -= g.Ho;
-      n += 1;
-    }
-  };
-#pragma unroll
-  for (int q = 0; q < 2; ++q)
-    init_cursor(np0 + snp + 32 * q, pnA[q], phoA[q], pwoA[q]);
-#pragma unroll
-  for (int q = 0; q < QB; ++q)
-    init_cursor(np0 + snpb + (64 / QB) * q, pnB[q], phoB[q], pwoB[q]);
+    pho[q] = rem / g.Wo;
+    pwo[q] = rem % g.Wo;
+  }
 
-  f32x4 acc[2][FJ] = {};
+  f32x4 acc[2][2] = {};
 
   for (long p0 = np0; p0 < np1; p0 += WBK) {
-    __bf16 ra[2][8], rb[QB][8];
+    __bf16 ra[2][8], rb[2][8];
 #pragma unroll
     for (int q = 0; q < 2; ++q) {
       const long m = p0 + snp + 32 * q;
@@ -432,22 +404,14 @@ __global__ __launch_bounds__(256) void conv_wgrad_fast_kernel(
       if (v && a_ok) {
         *reinterpret_cast<s16x8*>(ra[q]) = *reinterpret_cast<const s16x8*>(
             dy + m * g.K + k0c + scol);
-      } else if (v) {
-        row8(dy, NP, g.K, m, k0c + scol, ra[q]);
       } else {
         zero8(ra[q]);
       }
-      if (v) adv_cursor(phoA[q], pwoA[q], pnA[q]);
-    }
-#pragma unroll
-    for (int q = 0; q < QB; ++q) {
-      const long m = p0 + snpb + (64 / QB) * q;
-      const bool v = m < np1;
-      if (v && b_vec_ok) {
-        int h = phoB[q] * g.stride - g.pad + fb_r;
-        int w = pwoB[q] * g.stride - g.pad + fb_s;
+      if (v && bcol_ok) {
+        int h = pho[q] * g.stride - g.pad + fb_r;
+        int w = pwo[q] * g.stride - g.pad + fb_s;
         if ((unsigned)h < (unsigned)g.H && (unsigned)w < (unsigned)g.W) {
-          long off = (((long)pnB[q] * g.H + h) * g.W + w) * g.C + fb_c;
+          long off = (((long)pn[q] * g.H + h) * g.W + w) * g.C + fb_c;
           *reinterpret_cast<s16x8*>(rb[q]) =
               *reinterpret_cast<const s16x8*>(x + off);
         } else {
@@ -456,18 +420,25 @@ __global__ __launch_bounds__(256) void conv_wgrad_fast_kernel(
       } else {
         zero8(rb[q]);
       }
-      if (v) adv_cursor(phoB[q], pwoB[q], pnB[q]);
+      if (v) {
+        pwo[q] += WBK;
+        while (pwo[q] >= g.Wo) {
+          pwo[q] -= g.Wo;
+          pho[q] += 1;
+        }
+        while (pho[q] >= g.Ho) {
+          pho[q] -= g.Ho;
+          pn[q] += 1;
+        }
+      }
     }
     __syncthreads();
 #pragma unroll
     for (int q = 0; q < 2; ++q) {
-      *reinterpret_cast<bf16x8*>(&sA[simg + np_img(snp + 32 * q)]) =
+      const int npl = snp + 32 * q;
+      *reinterpret_cast<bf16x8*>(&sA[simg + np_img(npl)]) =
           *reinterpret_cast<bf16x8*>(ra[q]);
-    }
-#pragma unroll
-    for (int q = 0; q < QB; ++q) {
-      *reinterpret_cast<bf16x8*>(
-          &sB[simgb + np_img(snpb + (64 / QB) * q)]) =
+      *reinterpret_cast<bf16x8*>(&sB[simg + np_img(npl)]) =
           *reinterpret_cast<bf16x8*>(rb[q]);
     }
     __syncthreads();
@@ -479,9 +450,9 @@ __global__ __launch_bounds__(256) void conv_wgrad_fast_kernel(
         bf16x8 af =
             tr16_frag(&sA[(wr * 2 + fi) * WIMG], ks * 32, lane);
 #pragma unroll
-        for (int fj = 0; fj < FJ; ++fj) {
+        for (int fj = 0; fj < 2; ++fj) {
           bf16x8 bfr =
-              tr16_frag(&sB[(wc * FJ + fj) * WIMG], ks * 32, lane);
+              tr16_frag(&sB[(wc * 2 + fj) * WIMG], ks * 32, lane);
           acc[fi][fj] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               af, bfr, acc[fi][fj], 0, 0, 0);
         }
@@ -494,8 +465,8 @@ __global__ __launch_bounds__(256) void conv_wgrad_fast_kernel(
 #pragma unroll
   for (int fi = 0; fi < 2; ++fi) {
 #pragma unroll
-    for (int fj = 0; fj < FJ; ++fj) {
-      long nn = n0 + wc * (TBN / 2) + fj * 16 + cn;
+    for (int fj = 0; fj < 2; ++fj) {
+      long nn = n0 + wc * 32 + fj * 16 + cn;
       if (nn >= RED) continue;
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
@@ -791,32 +762,32 @@ torch::Tensor conv2d_wgrad(torch::Tensor x, torch::Tensor dy, long R, long S,
               "wgrad geometry mismatch");
   long NP = (long)g.N * g.Ho * g.Wo;
   long RED = (long)R * S * g.C;
-  // split the NP reduction across blocks for parallelism; fp32 atomics.
-  // wide (128-col) rsc tiles halve the dy/x re-staging for large RED.
-  bool fast = (g.K % 8 == 0) && (g.C % 8 == 0);
-  long tbn = (fast && RED >= 128) ? 128 : 64;
+  // split the NP reduction across blocks for parallelism; fp32 atomics
   long target_blocks = 2048;
-  long tiles = (long)ceil_div(g.K, conv::WBM) * ceil_div(RED, tbn);
+  long tiles = (long)ceil_div(g.K, conv::WBM) * ceil_div(RED, conv::WBN);
   long zsplit =
       std::max(1L, std::min(512L, target_blocks / std::max(tiles, 1L)));
   long npslice = (NP + zsplit - 1) / zsplit;
   npslice = ((npslice + conv::WBK - 1) / conv::WBK) * conv::WBK;
   zsplit = (NP + npslice - 1) / npslice;
+  bool fast = (g.K % 8 == 0) && (g.C % 8 == 0);
   // single NP slice per tile -> plain stores, no zero-init needed
-  auto dw = (fast && zsplit == 1)
+  auto dw = zsplit == 1
       ? torch::empty({(long)g.K, RED}, x.options().dtype(torch::kFloat32))
       : torch::zeros({(long)g.K, RED}, x.options().dtype(torch::kFloat32));
-  dim3 grid(ceil_div(g.K, conv::WBM), ceil_div(RED, tbn), (unsigned)zsplit);
-  if (fast) {
-    auto* fn = zsplit == 1
-        ? (tbn == 128 ? conv::conv_wgrad_fast_kernel<false, 128>
-                      : conv::conv_wgrad_fast_kernel<false, 64>)
-        : (tbn == 128 ? conv::conv_wgrad_fast_kernel<true, 128>
-                      : conv::conv_wgrad_fast_kernel<true, 64>);
-    fn<<<grid, 256, 0, cur_stream()>>>(
+  dim3 grid(ceil_div(g.K, conv::WBM), ceil_div(RED, conv::WBN),
+            (unsigned)zsplit);
+  if (fast && zsplit == 1) {
+    conv::conv_wgrad_fast_kernel<false><<<grid, 256, 0, cur_stream()>>>(
+        (const bf16*)x.data_ptr(), (const bf16*)dy.data_ptr(),
+        dw.data_ptr<float>(), g, NP, npslice);
+  } else if (fast) {
+    conv::conv_wgrad_fast_kernel<true><<<grid, 256, 0, cur_stream()>>>(
         (const bf16*)x.data_ptr(), (const bf16*)dy.data_ptr(),
         dw.data_ptr<float>(), g, NP, npslice);
   } else {
+    TORCH_CHECK(zsplit >= 1);
+    if (zsplit == 1) dw.zero_();
     conv::conv_wgrad_kernel<false><<<grid, 256, 0, cur_stream()>>>(
         (const bf16*)x.data_ptr(), (const bf16*)dy.data_ptr(),
         dw.data_ptr<float>(), g, NP, npslice);
