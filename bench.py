@@ -64,7 +64,9 @@ def main() -> int:
     # synthetic data, resident batches (data=synthetic per BASELINE contract)
     g = torch.Generator(device="cpu").manual_seed(1234 + rank)
     n_resident = 8
-    xs = [torch.randn(args.batch, 3, 32, 32, generator=g).to(device)
+    shape = ((1, 28, 28) if args.model in ("cnn1", "cnn2", "mlp")
+             else (3, 32, 32))
+    xs = [torch.randn(args.batch, *shape, generator=g).to(device)
           for _ in range(n_resident)]
     ys = [torch.randint(0, 10, (args.batch,), generator=g).to(device)
           for _ in range(n_resident)]
@@ -127,8 +129,9 @@ def main() -> int:
             "dtype": "bf16",
             "data": "synthetic",
             "config": {
-                "model": "resnet18-quirk (ref ResNet<BasicBlock>{2,2,2,2}, "
-                         "17444682 params / 86 tensors)",
+                "model": ("resnet18-quirk (ref ResNet<BasicBlock>{2,2,2,2}, "
+                          "17444682 params / 86 tensors)"
+                          if args.model == "resnet18q" else args.model),
                 "mode": args.mode,
                 "global_batch": world * args.batch,
                 "per_gpu_batch": args.batch,
