@@ -19,7 +19,7 @@ dcifar10/event/event.cpp:299-365 and spevent.cpp:321-426):
         num_events += 2                  # one per neighbor
 
 The controller state is tiny (sz <= 86 scalars per array) — on GPU the same
-update runs device-resident in the HIP trigger kernel (csrc/sgd_trigger.hip)
+update runs device-resident in the HIP trigger kernel (csrc/engine.hip trigger_update)
 and this class is used for CPU runs, unit tests, and kernel parity checks.
 """
 

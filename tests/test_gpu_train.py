@@ -151,3 +151,31 @@ def test_hipgraph_step_matches_eager():
     _, loss2 = graph.step(x2, y)
     torch.cuda.synchronize()
     assert abs(float(loss2.detach()) - lg) > 1e-6  # input actually flows
+
+
+def test_gpu_checkpoint_roundtrip(tmp_path):
+    """Checkpoint/resume with the device-resident controller state."""
+    import os
+    from eventgrad_amd.config import DataConfig, OptimConfig, RunConfig
+    from eventgrad_amd.train.trainer import Trainer
+
+    ck = os.path.join(tmp_path, "ck.pt")
+
+    def cfg(epochs, resume):
+        c = RunConfig(
+            mode="serial", model="cnn2", epochs=epochs, device="cuda",
+            data=DataConfig(dataset="synthetic-mnist", batch_size=64,
+                            synthetic_train_samples=256,
+                            synthetic_test_samples=64),
+            optim=OptimConfig(lr=0.05, momentum=0.9), eval_at_end=False,
+            checkpoint_path=ck, resume=resume, final_consensus=False)
+        return c
+
+    tr1 = Trainer(cfg(1, False))
+    tr1.train()
+    p_after1 = tr1.space.param.clone()
+    tr2 = Trainer(cfg(2, True))
+    assert torch.allclose(tr2.space.param, p_after1)
+    m = tr2.train()
+    assert tr2.pass_num == 8  # 4 passes/epoch x 2 epochs total
+    assert np.isfinite(m.final_train_loss)
