@@ -347,8 +347,10 @@ std::vector<torch::Tensor> bn_fwd(torch::Tensor x, torch::Tensor gamma,
     if (have_stats) {
       // sums already accumulated by the conv epilogue
     } else if (c % 8 == 0 && c <= bn::BN_MAXC) {
+      // block cap 256: measured sweet spot between per-thread TLP and the
+      // per-channel global-atomic merge (128/512/1024/4096 all slower)
       int rpb = 256 / (c / 8);
-      int grid = (int)std::min<long>((rows + rpb - 1) / rpb, 1024L);
+      int grid = (int)std::min<long>((rows + rpb - 1) / rpb, 256L);
       bn::stats_kernel_v<<<grid, 256, 0, stream>>>(
           (const bf16*)x.data_ptr(), s.data_ptr<float>(),
           sq.data_ptr<float>(), rows, c);
@@ -404,7 +406,7 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor x,
   auto stream = cur_stream();
   if (c % 8 == 0 && c <= bn::BN_MAXC) {
     int rpb = 256 / (c / 8);
-    int grid = (int)std::min<long>((rows + rpb - 1) / rpb, 1024L);
+    int grid = (int)std::min<long>((rows + rpb - 1) / rpb, 256L);
     bn::bwd_stats_kernel_v<<<grid, 256, 0, stream>>>(
         (const bf16*)dy.data_ptr(), (const bf16*)x.data_ptr(),
         (const bf16*)y.data_ptr(), mean.data_ptr<float>(),
@@ -421,7 +423,7 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor x,
   auto dx = torch::empty_like(dy);
   if (c % 8 == 0) {
     int rpb = 256 / (c / 8);
-    int grid1 = (int)std::min<long>((rows + rpb - 1) / rpb, 1024L);
+    int grid1 = (int)std::min<long>((rows + rpb - 1) / rpb, 256L);
     bn::bwd_dx_kernel_v<<<grid1, 256, 0, stream>>>(
         (const bf16*)dy.data_ptr(), (const bf16*)x.data_ptr(),
         (const bf16*)y.data_ptr(), mean.data_ptr<float>(),
