@@ -54,6 +54,11 @@ torch::Tensor logsoftmax_nll_bwd(torch::Tensor, torch::Tensor, torch::Tensor);
 // topk.hip
 std::vector<torch::Tensor> topk_absdiff(torch::Tensor, torch::Tensor, long);
 void scatter_update(torch::Tensor, torch::Tensor, torch::Tensor);
+torch::Tensor spevent_pack(torch::Tensor, torch::Tensor, torch::Tensor,
+                           torch::Tensor, torch::Tensor, torch::Tensor, long,
+                           long);
+void spevent_unpack(torch::Tensor, torch::Tensor, torch::Tensor,
+                    torch::Tensor, torch::Tensor, long);
 torch::Tensor tr16_probe(long);
 }  // namespace eg
 
@@ -92,5 +97,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("logsoftmax_nll_bwd", &eg::logsoftmax_nll_bwd);
   m.def("topk_absdiff", &eg::topk_absdiff);
   m.def("scatter_update", &eg::scatter_update);
+  m.def("spevent_pack", &eg::spevent_pack);
+  m.def("spevent_unpack", &eg::spevent_unpack);
   m.def("tr16_probe", &eg::tr16_probe);
 }

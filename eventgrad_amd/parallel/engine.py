@@ -329,24 +329,52 @@ class SparseGossipEngine(GossipEngine):
     def _payload_elems(self, fired_idx):
         return sum(2 * self.k[i] for i in fired_idx)
 
+    def _seg_tables(self, fired_idx):
+        """Device tables for the batched pack/unpack kernels.
+
+        offs layout: [val_offs[0..nf-1], total_payload,
+                      cum_k[0..nf-1], total_k]  (see csrc/topk.hip).
+        """
+        dev = self.device
+        val_offs, cum_k = [], []
+        off = ck = 0
+        for i in fired_idx:
+            val_offs.append(off)
+            cum_k.append(ck)
+            off += 2 * self.k[i]
+            ck += self.k[i]
+        starts = torch.tensor([self.space.starts[i] for i in fired_idx],
+                              dtype=torch.int64, device=dev)
+        lens = torch.tensor([self.space.numels[i] for i in fired_idx],
+                            dtype=torch.int64, device=dev)
+        ks = torch.tensor([self.k[i] for i in fired_idx],
+                          dtype=torch.int64, device=dev)
+        offs = torch.tensor(val_offs + [off] + cum_k + [ck],
+                            dtype=torch.int64, device=dev)
+        return starts, lens, ks, offs, off, ck
+
     def _make_send(self, fired_idx):
         if not fired_idx:
             e = self.space.param.new_empty(0)
             return e, e
+        if self.device.type == "cuda":
+            # one fused 9-launch radix-select for ALL fired tensors; the
+            # kernel writes the wire payload and updates prev in place
+            starts, lens, ks, offs, total, _ = self._seg_tables(fired_idx)
+            payload = native().spevent_pack(
+                self.space.param, self.prev, starts, lens, ks, offs, total,
+                max(self.space.numels[i] for i in fired_idx))
+            return payload, payload.clone()
         chunks = []
         for i in fired_idx:
             seg = self.space.seg(self.space.param, i)
             pseg = self.space.seg(self.prev, i)
-            if self.device.type == "cuda":
-                # kernel also updates prev[idx] = vals (spevent.cpp:407-413)
-                vals, idx = native().topk_absdiff(seg, pseg, self.k[i])
-            else:
-                diff = (seg - pseg).abs()
-                _, idx = torch.topk(diff, self.k[i], sorted=True)
-                vals = seg[idx]
-                # update prev at the sent indices only (spevent.cpp:407-413)
-                pseg[idx] = vals
-                idx = idx.to(torch.int32)
+            diff = (seg - pseg).abs()
+            _, idx = torch.topk(diff, self.k[i], sorted=True)
+            vals = seg[idx]
+            # update prev at the sent indices only (spevent.cpp:407-413)
+            pseg[idx] = vals
+            idx = idx.to(torch.int32)
             chunks.append(vals.to(torch.float32))
             chunks.append(idx.view(torch.float32))
         payload = torch.cat(chunks)
@@ -357,16 +385,19 @@ class SparseGossipEngine(GossipEngine):
                            dtype=torch.float32, device=self.device)
 
     def _apply_recv(self, payload, fired_idx, replica):
+        if not fired_idx:
+            return
+        if self.device.type == "cuda":
+            starts, lens, ks, offs, _, total_k = self._seg_tables(fired_idx)
+            native().spevent_unpack(payload, starts, ks, offs, replica,
+                                    total_k)
+            return
         off = 0
         for i in fired_idx:
             k = self.k[i]
             vals = payload[off:off + k]
             idx = payload[off + k:off + 2 * k].view(torch.int32)
-            seg = self.space.seg(replica, i)
-            if self.device.type == "cuda":
-                native().scatter_update(seg, vals, idx)
-            else:
-                seg[idx.long()] = vals
+            self.space.seg(replica, i)[idx.long()] = vals
             off += 2 * k
 
 

@@ -367,3 +367,61 @@ def test_conv_fused_bn_stats_matches_separate():
     # means are near zero so compare absolutely at bf16-noise scale
     assert torch.allclose(rm2, rm1, atol=3e-4), (rm2 - rm1).abs().max()
     assert rel_err(rv2, rv1) < 1e-2
+
+
+def test_spevent_pack_unpack_batched():
+    """Fused multi-tensor pack/unpack vs the single-tensor kernels."""
+    torch.manual_seed(33)
+    dev = "cuda"
+    # three segments in one flat buffer, 64-aligned starts
+    numels = [1000, 257, 4096]
+    starts, off = [], 0
+    for n in numels:
+        starts.append(off)
+        off += (n + 63) // 64 * 64
+    flat = torch.zeros(off, device=dev)
+    prev = torch.zeros(off, device=dev)
+    for s, n in zip(starts, numels):
+        flat[s:s + n] = torch.randn(n, device=dev)
+        prev[s:s + n] = torch.randn(n, device=dev)
+    prev0 = prev.clone()
+    ks = [50, 17, 123]
+    val_offs, cum_k, o, ck = [], [], 0, 0
+    for k in ks:
+        val_offs.append(o)
+        cum_k.append(ck)
+        o += 2 * k
+        ck += k
+    t64 = lambda v: torch.tensor(v, dtype=torch.int64, device=dev)
+    payload = core().spevent_pack(flat, prev, t64(starts), t64(numels),
+                                  t64(ks), t64(val_offs + [o] + cum_k + [ck]),
+                                  o, max(numels))
+    # reference: per-segment torch.topk of |flat - prev0|
+    off2 = 0
+    for (s, n, k) in zip(starts, numels, ks):
+        seg, pseg = flat[s:s + n], prev0[s:s + n]
+        ref_vals, _ = torch.topk((seg - pseg).abs(), k)
+        got_vals = payload[off2:off2 + k]
+        got_idx = payload[off2 + k:off2 + 2 * k].view(torch.int32).long()
+        # same selected |diff| multiset; vals are flat at the indices
+        got_diff = (seg - pseg).abs()[got_idx].sort().values
+        assert torch.allclose(got_diff, ref_vals.sort().values, atol=1e-6)
+        assert torch.allclose(got_vals, seg[got_idx])
+        # prev updated exactly at the selected indices
+        assert torch.allclose(prev[s:s + n][got_idx], seg[got_idx])
+        mask = torch.ones(n, dtype=torch.bool, device=dev)
+        mask[got_idx] = False
+        assert torch.equal(prev[s:s + n][mask], pseg[mask])
+        off2 += 2 * k
+    # unpack scatters into a replica exactly like per-segment scatter
+    replica = torch.randn(off, device=dev)
+    ref_replica = replica.clone()
+    core().spevent_unpack(payload, t64(starts), t64(ks),
+                          t64(val_offs + [o] + cum_k + [ck]), replica, ck)
+    off2 = 0
+    for (s, n, k) in zip(starts, numels, ks):
+        vals = payload[off2:off2 + k]
+        idx = payload[off2 + k:off2 + 2 * k].view(torch.int32).long()
+        ref_replica[s:s + n][idx] = vals
+        off2 += 2 * k
+    assert torch.equal(replica, ref_replica)
