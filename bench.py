@@ -77,6 +77,10 @@ def main() -> int:
         from eventgrad_amd.train.graphstep import FwdBwdGraph, can_graph
         if can_graph(model, device):
             graph = FwdBwdGraph(model, space, tuple(xs[0].shape), device)
+            # capture BEFORE any RCCL p2p is in flight (graph capture with
+            # outstanding comm on other streams is the risky combination)
+            graph.step(xs[0], ys[0])
+            torch.cuda.synchronize()
 
     def step():
         nonlocal pass_num

@@ -178,7 +178,9 @@ class CommEngine:
         if self.world > 1:
             dist.all_reduce(self.space.param)
             self.space.param.div_(self.world)
-            ev = torch.tensor([self.num_events], dtype=torch.int64)
+            # NCCL needs a device tensor (gloo accepts either)
+            ev = torch.tensor([self.num_events], dtype=torch.int64,
+                              device=self.device)
             dist.all_reduce(ev)
             total_events = int(ev.item())
         else:

@@ -78,11 +78,14 @@ class Trainer:
                 x = x.to(self.device, non_blocking=True)
                 y = y.to(self.device, non_blocking=True)
                 self.pass_num += 1
+                if use_graph and graph is None:
+                    # capture before the first comm pass posts RCCL work
+                    graph = FwdBwdGraph(self.model, self.space,
+                                        tuple(x.shape), self.device)
+                    graph.step(x, y)
+                    torch.cuda.synchronize()
                 self.engine.begin_pass(self.pass_num)
                 if use_graph:
-                    if graph is None:
-                        graph = FwdBwdGraph(self.model, self.space,
-                                            tuple(x.shape), self.device)
                     logits, loss = graph.step(x, y)
                 else:
                     self.space.zero_grad()
