@@ -122,13 +122,15 @@ __global__ __launch_bounds__(TBM * 4) void conv_mm_kernel(
     float* __restrict__ stats_ws = nullptr) {
   constexpr int THREADS = TBM * 4;
   constexpr int BROWS = 512 / THREADS;  // B-staging rows per thread
-  __shared__ __bf16 sA[TBM * CLDK];
-  __shared__ __bf16 sB[CBN * CLDK];
-  // per-block BN-stats accumulators (fused batch-norm statistics: the
-  // separate per-channel stats pass over y disappears when the consumer
-  // is a training-mode BatchNorm)
-  __shared__ float s_sum[STATS ? CBN : 1];
-  __shared__ float s_sq[STATS ? CBN : 1];
+  // ONE shared allocation (guide §5.4 trap 4a: a second __shared__ object
+  // makes hipcc over-synchronize the k-loop — measured +14 waitcnts and an
+  // extra barrier per step with separate stats arrays). Stats accumulators
+  // live in a float-aliased tail of the same array.
+  __shared__ __bf16 smem[TBM * CLDK + CBN * CLDK + (STATS ? 2 * CBN * 2 : 0)];
+  __bf16* sA = smem;
+  __bf16* sB = smem + TBM * CLDK;
+  float* s_sum = reinterpret_cast<float*>(smem + TBM * CLDK + CBN * CLDK);
+  float* s_sq = s_sum + CBN;
   if (STATS && threadIdx.x < CBN) {
     s_sum[threadIdx.x] = 0.f;
     s_sq[threadIdx.x] = 0.f;

@@ -90,6 +90,7 @@ class RunConfig:
     trace: bool = False           # write send{rank}.txt / recv{rank}.txt
     trace_dir: str = "."
     log_interval: int = 20        # dcifar10 Options::log_interval
+    #                               (defined-but-unused in the reference too)
     # checkpointing (new capability; reference has none — SURVEY.md §5)
     checkpoint_path: Optional[str] = None
     checkpoint_every_epochs: int = 0   # 0 = only at end if path set
