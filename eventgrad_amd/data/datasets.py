@@ -6,7 +6,7 @@ import gzip
 import os
 import pickle
 import struct
-from typing import Optional, Tuple
+from typing import Tuple
 
 import numpy as np
 import torch

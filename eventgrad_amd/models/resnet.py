@@ -18,7 +18,6 @@ quoted on) and also provide standard counts (``resnet18``).
 
 from __future__ import annotations
 
-import torch
 from torch import nn
 
 from ..ops import functional as O
