@@ -425,3 +425,53 @@ def test_spevent_pack_unpack_batched():
         ref_replica[s:s + n][idx] = vals
         off2 += 2 * k
     assert torch.equal(replica, ref_replica)
+
+
+def test_conv2d_fuzz_shapes():
+    """Seeded fuzz: random conv geometries vs the fp32 oracle."""
+    rng = np.random.default_rng(7)
+    from eventgrad_amd.ops import functional as O
+    for trial in range(10):
+        N = int(rng.integers(1, 5))
+        C = int(rng.choice([3, 8, 16, 24, 40, 64, 96]))
+        K = int(rng.choice([8, 16, 32, 48, 64, 128]))
+        R = int(rng.choice([1, 3, 5]))
+        H = int(rng.integers(R, 20))
+        stride = int(rng.choice([1, 2]))
+        pad = int(rng.integers(0, (R + 1) // 2 + 1))
+        if (H + 2 * pad - R) // stride + 1 < 1:
+            continue
+        torch.manual_seed(100 + trial)
+        x_f = torch.randn(N, C, H, H)
+        w_f = torch.randn(K, C, R, R) * (0.5 / (R * np.sqrt(C)))
+        x_g = bq(x_f).to(DEV).permute(0, 2, 3, 1).contiguous() \
+            .to(torch.bfloat16).requires_grad_(True)
+        w_g = w_f.clone().to(DEV).requires_grad_(True)
+        y_g = O.conv2d(x_g, w_g, None, stride, pad)
+        x_c = bq(x_f).requires_grad_(True)
+        w_c = bq(w_f).requires_grad_(True)
+        y_c = F.conv2d(x_c, w_c, None, stride=stride, padding=pad)
+        tag = f"trial{trial} N{N} C{C} K{K} R{R} H{H} s{stride} p{pad}"
+        assert rel_err(y_g.float().permute(0, 3, 1, 2).cpu(),
+                       y_c.detach()) < 4e-2, ("fwd", tag)
+        dy = torch.randn_like(y_c)
+        y_c.backward(dy)
+        y_g.backward(bq(dy).to(DEV).permute(0, 2, 3, 1).contiguous()
+                     .to(torch.bfloat16))
+        assert rel_err(w_g.grad.cpu(), w_c.grad) < 4e-2, ("wgrad", tag)
+        assert rel_err(x_g.grad.float().permute(0, 3, 1, 2).cpu(),
+                       x_c.grad) < 4e-2, ("dgrad", tag)
+
+
+def test_gemm_fuzz_shapes():
+    rng = np.random.default_rng(11)
+    for trial in range(10):
+        M = int(rng.integers(1, 300))
+        N = int(rng.integers(1, 300))
+        K = int(rng.integers(1, 600))
+        torch.manual_seed(200 + trial)
+        A = torch.randn(M, K, device=DEV).to(torch.bfloat16)
+        B = torch.randn(N, K, device=DEV).to(torch.bfloat16)
+        C = core().gemm_bias(A, B, torch.empty(0, device=DEV), False)
+        ref = bq(A.float()).cpu() @ bq(B.float()).cpu().t()
+        assert rel_err(C.cpu(), ref) < 3e-2, (trial, M, N, K)
