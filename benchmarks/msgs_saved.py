@@ -52,6 +52,8 @@ def build_cfg(experiment: str, mode: str, args):
     cfg.trigger.initial_comm_passes = args.warmup_passes
     cfg.topk_percent = args.topk_percent
     cfg.device = args.device
+    if args.backend:
+        cfg.dist_backend = args.backend
     return cfg
 
 
@@ -94,6 +96,9 @@ def main():
     ap.add_argument("--warmup-passes", type=int, default=30)
     ap.add_argument("--topk-percent", type=float, default=1.0)
     ap.add_argument("--device", default="cpu")
+    ap.add_argument("--backend", default=None,
+                    help="force dist backend (e.g. gloo for N ranks on one "
+                         "GPU; the wire stages through host)")
     ap.add_argument("--modes", nargs="+",
                     default=["decent", "event"])
     ap.add_argument("--outdir", default="benchmarks/out")

@@ -97,6 +97,7 @@ class RunConfig:
     resume: bool = False
     # device/dtype
     device: str = "auto"          # auto | cpu | cuda
+    dist_backend: Optional[str] = None  # override (nccl/gloo); None = auto
     compute_dtype: str = "bf16"   # bf16 | fp32 (GPU compute dtype; params fp32)
     hip_graph: bool = True        # capture fwd+bwd into a hipGraph on GPU
     #                               (auto-skipped for models with dropout)

@@ -29,7 +29,8 @@ from .trace import Tracer
 class Trainer:
     def __init__(self, cfg: RunConfig):
         self.cfg = cfg
-        self.rank, self.world, self.device = init_distributed(cfg.device)
+        self.rank, self.world, self.device = init_distributed(
+            cfg.device, backend=cfg.dist_backend)
         torch.manual_seed(cfg.seed)  # ref: torch::manual_seed(0) everywhere
 
         self.model = build_model(cfg.model, cfg.data.num_classes)
