@@ -43,6 +43,9 @@ def main() -> int:
     ap.add_argument("--model", default="resnet18q")
     ap.add_argument("--no-graph", action="store_true",
                     help="disable hipGraph capture of fwd+bwd")
+    ap.add_argument("--backend", default=None,
+                    help="dist backend override (testing: gloo lets two "
+                         "ranks share one GPU; default nccl=RCCL)")
     args = ap.parse_args()
 
     cfg = preset("dcifar10-event")
@@ -51,7 +54,7 @@ def main() -> int:
     cfg.data.global_batch = None
     cfg.data.batch_size = args.batch
 
-    rank, world, device = init_distributed("auto")
+    rank, world, device = init_distributed("auto", backend=args.backend)
     if device.type != "cuda":
         raise SystemExit("bench.py requires a GPU")
     torch.manual_seed(cfg.seed)
