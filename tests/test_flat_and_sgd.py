@@ -81,3 +81,24 @@ def test_avg3_semantics():
     before = space.param.clone()
     _CpuK.avg3(space.param, left, right)
     assert torch.allclose(space.param, (before + 6) / 3)
+
+
+def test_sgd_weight_decay_matches_torch():
+    torch.manual_seed(0)
+    m1 = build_model("mlp")
+    torch.manual_seed(0)
+    m2 = build_model("mlp")
+    space = FlatParamSpace(m1, torch.device("cpu"))
+    opt = torch.optim.SGD(m2.parameters(), lr=0.05, momentum=0.9,
+                          weight_decay=1e-4)
+    for it in range(4):
+        torch.manual_seed(300 + it)
+        x = torch.randn(8, 1, 28, 28)
+        space.zero_grad()
+        m1(x).square().mean().backward()
+        opt.zero_grad()
+        m2(x).square().mean().backward()
+        _CpuK.sgd_step(space, lr=0.05, momentum=0.9, wd=1e-4)
+        opt.step()
+        for p1, p2 in zip(m1.parameters(), m2.parameters()):
+            assert torch.allclose(p1, p2, atol=1e-7)

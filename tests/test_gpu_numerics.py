@@ -475,3 +475,20 @@ def test_gemm_fuzz_shapes():
         C = core().gemm_bias(A, B, torch.empty(0, device=DEV), False)
         ref = bq(A.float()).cpu() @ bq(B.float()).cpu().t()
         assert rel_err(C.cpu(), ref) < 3e-2, (trial, M, N, K)
+
+
+def test_sgd_step_norm_weight_decay():
+    flat, st, nu, starts, numels = _mk_flat(seed=17)
+    mask = torch.zeros_like(flat)
+    for s, n in zip(starts, numels):
+        mask[s:s + n] = 1
+    grad = torch.randn_like(flat) * mask
+    mom = torch.randn_like(flat) * mask
+    p_ref, g_ref, m_ref = flat.cpu().clone(), grad.cpu(), mom.cpu().clone()
+    core().sgd_step_norm(flat, grad, mom, st, nu, 0.05, 0.9, 1e-4)
+    g_eff = g_ref + 1e-4 * p_ref
+    m_ref.mul_(0.9).add_(g_eff)
+    p_ref.add_(m_ref, alpha=-0.05)
+    # pad gaps get wd*0 contributions only where param is 0 -> identical
+    assert rel_err(flat.cpu() * mask.cpu(), p_ref * mask.cpu()) < 1e-6
+    assert rel_err(mom.cpu() * mask.cpu(), m_ref * mask.cpu()) < 1e-6
