@@ -6,7 +6,7 @@ import os
 
 import torch.multiprocessing as mp
 
-_NEXT_PORT = [29600 + (os.getpid() % 100) * 3]
+_NEXT_PORT = [29100 + (os.getpid() % 100) * 3]  # disjoint from benchmarks (29650+)
 
 
 def next_port() -> int:
