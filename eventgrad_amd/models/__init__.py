@@ -35,4 +35,5 @@ def build_model(name: str, num_classes: int = 10):
 
 MODEL_NAMES = ("mlp", "cnn1", "cnn2", "lenet5", "resnet18q", "resnet18",
                "resnet34q", "resnet34", "resnet50q", "resnet50",
-               "resnet101", "resnet152")
+               "resnet101", "resnet152", "resnet20", "resnet32",
+               "resnet44", "resnet56")

@@ -11,6 +11,7 @@ from eventgrad_amd.models import build_model
     ("cnn2", 8, 27480),        # event.cpp CNN-2
     ("resnet18q", 86, 17444682),  # quirk ResNet
     ("resnet18", 62, 11173962),   # standard ResNet-18 (CIFAR stem)
+    ("resnet20", 65, 272474),     # classic CIFAR ResNet-20 (He et al.)
 ])
 def test_param_counts(name, n_tensors, n_params):
     m = build_model(name)
@@ -26,6 +27,7 @@ def test_param_counts(name, n_tensors, n_params):
     ("lenet5", (2, 3, 32, 32)),
     ("resnet18q", (2, 3, 32, 32)),
     ("resnet50q", (2, 3, 32, 32)),
+    ("resnet20", (2, 3, 32, 32)),
 ])
 def test_forward_shapes(name, shape):
     torch.manual_seed(0)
