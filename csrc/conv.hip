@@ -423,15 +423,20 @@ __global__ __launch_bounds__(256) void conv_wgrad_fast_kernel(
         zero8(rb[q]);
       }
       if (v) {
-        pwo[q] += WBK;
-        while (pwo[q] >= g.Wo) {
+        // constant-delta advance (see dwo/dho/dn above): <= 2 cond. wraps
+        pwo[q] += dwo;
+        int car = 0;
+        if (pwo[q] >= g.Wo) {
           pwo[q] -= g.Wo;
-          pho[q] += 1;
+          car = 1;
         }
-        while (pho[q] >= g.Ho) {
+        pho[q] += dho + car;
+        int car2 = 0;
+        if (pho[q] >= g.Ho) {
           pho[q] -= g.Ho;
-          pn[q] += 1;
+          car2 = 1;
         }
+        pn[q] += dn + car2;
       }
     }
     __syncthreads();
