@@ -54,11 +54,14 @@ def event_equals_decent_worker(rank, world, port, outdir):
     from eventgrad_amd.train.trainer import Trainer
 
     params = {}
-    for mode, trig in [
+    for cycle, (mode, trig) in enumerate([
         ("decent", TriggerConfig()),
         ("event", TriggerConfig(adaptive=False, constant=0.0,
                                 initial_comm_passes=0)),
-    ]:
+    ]):
+        # fresh port per init cycle: re-binding the TCPStore on the same
+        # port right after destroy_process_group races (observed hangs)
+        os.environ["MASTER_PORT"] = str(port + 1 + cycle)
         cfg = _small_cfg(mode, trigger=trig, epochs=2)
         tr = Trainer(cfg)
         tr.train()
@@ -104,7 +107,9 @@ def checkpoint_resume_worker(rank, world, port, outdir):
     torch.distributed.destroy_process_group()
 
     # intermediate run must not run the closing consensus allreduce, so the
-    # checkpoint holds exactly the straight run's end-of-epoch-1 state
+    # checkpoint holds exactly the straight run's end-of-epoch-1 state.
+    # Fresh port per init cycle (TCPStore rebind race -> hangs).
+    os.environ["MASTER_PORT"] = str(port + 1)
     cfg1 = _small_cfg("event", epochs=1, momentum=0.9)
     cfg1.checkpoint_path = ck
     cfg1.final_consensus = False
@@ -112,6 +117,7 @@ def checkpoint_resume_worker(rank, world, port, outdir):
     tr1.train()
     torch.distributed.destroy_process_group()
 
+    os.environ["MASTER_PORT"] = str(port + 2)
     cfg2 = _small_cfg("event", epochs=2, momentum=0.9)
     cfg2.checkpoint_path = ck
     cfg2.resume = True

@@ -10,7 +10,7 @@ _NEXT_PORT = [29100 + (os.getpid() % 100) * 3]  # disjoint from benchmarks (2965
 
 
 def next_port() -> int:
-    _NEXT_PORT[0] += 1
+    _NEXT_PORT[0] += 3  # leave room for per-cycle +1/+2 offsets in workers
     return _NEXT_PORT[0]
 
 
