@@ -394,6 +394,13 @@ __global__ __launch_bounds__(256) void conv_wgrad_fast_kernel(
     pho[q] = rem / g.Wo;
     pwo[q] = rem % g.Wo;
   }
+  // O(1) cursor advance: np += WBK decomposes into per-shape constants
+  // (the while-wrap form costs up to ~16 serial VALU iterations per step
+  // for the deep stages' small Wo/Ho)
+  const int dwo = WBK % g.Wo;
+  const int dcar = WBK / g.Wo;
+  const int dho = dcar % g.Ho;
+  const int dn = dcar / g.Ho;
 
   f32x4 acc[2][2] = {};
 
