@@ -39,8 +39,11 @@ class Conv2d(nn.Module):
             nn.init.uniform_(self.bias, -bound, bound)
 
     def forward(self, x, bn_stats: bool = False):
+        # fused-BN-stats epilogue requires out_ch % 64 == 0 (csrc/conv.hip);
+        # the paired BatchNorm downgrades stats_ready by the same condition
         return O.conv2d(x, self.weight, self.bias, self.stride, self.padding,
-                        bn_stats=bn_stats and x.is_cuda and self.training)
+                        bn_stats=bn_stats and x.is_cuda and self.training
+                        and self.out_ch % 64 == 0)
 
     def extra_repr(self):
         return (f"{self.in_ch}, {self.out_ch}, k={self.kernel_size}, "
@@ -74,7 +77,8 @@ class BatchNorm2d(nn.Module):
         return O.batch_norm(x, self.weight, self.bias, self.running_mean,
                             self.running_var, self.training, self.momentum,
                             self.eps, fuse_relu,
-                            stats_ready and x.is_cuda and self.training)
+                            stats_ready and x.is_cuda and self.training
+                            and self.num_features % 64 == 0)
 
 
 class Linear(nn.Module):

@@ -181,6 +181,30 @@ def test_gpu_checkpoint_roundtrip(tmp_path):
     assert np.isfinite(m.final_train_loss)
 
 
+def test_resnet20_gpu_step():
+    """Classic CIFAR ResNet-20 on GPU (16/32-channel convs exercise the
+    fused-stats downgrade path: K % 64 != 0 -> separate BN stats)."""
+    from eventgrad_amd.models import build_model
+    from eventgrad_amd.ops import functional as O
+    from eventgrad_amd.parallel.flat import FlatParamSpace
+    from eventgrad_amd.ops.backend import native
+
+    dev = torch.device("cuda")
+    torch.manual_seed(0)
+    m = build_model("resnet20").to(dev)
+    m.train()
+    space = FlatParamSpace(m, dev)
+    x = torch.randn(8, 3, 32, 32, device=dev)
+    y = torch.randint(0, 10, (8,), device=dev)
+    loss = O.nll_of_logits(m(x), y)
+    loss.backward()
+    native().sgd_step_norm(space.param, space.grad, space.momentum,
+                           space.starts_t, space.numels_t, 0.01, 0.9, 0.0)
+    torch.cuda.synchronize()
+    assert torch.isfinite(loss).item()
+    assert bool(torch.isfinite(space.grad).all().item())
+
+
 def test_bottleneck_resnet_gpu_step():
     """BottleNeck (1-3-1) block path on GPU: one fwd/bwd/step, finite."""
     from eventgrad_amd.models import build_model
