@@ -228,3 +228,31 @@ def test_bottleneck_resnet_gpu_step():
     assert torch.isfinite(loss).item()
     assert bool(torch.isfinite(norms).all().item())
     assert bool(torch.isfinite(space.grad).all().item())
+
+
+@pytest.mark.parametrize("name", ["mlp", "cnn1", "cnn2", "lenet5",
+                                  "resnet18", "resnet34q", "resnet50",
+                                  "resnet32", "resnet101"])
+def test_model_zoo_gpu_step(name):
+    """Every zoo architecture: one fwd/bwd/step on GPU, finite results."""
+    from eventgrad_amd.models import build_model
+    from eventgrad_amd.ops import functional as O
+    from eventgrad_amd.parallel.flat import FlatParamSpace
+    from eventgrad_amd.ops.backend import native
+
+    dev = torch.device("cuda")
+    torch.manual_seed(0)
+    m = build_model(name).to(dev)
+    m.train()
+    space = FlatParamSpace(m, dev)
+    shape = (4, 1, 28, 28) if name in ("mlp", "cnn1", "cnn2") \
+        else (4, 3, 32, 32)
+    x = torch.randn(shape, device=dev)
+    y = torch.randint(0, 10, (4,), device=dev)
+    loss = O.nll_of_logits(m(x), y)
+    loss.backward()
+    native().sgd_step_norm(space.param, space.grad, space.momentum,
+                           space.starts_t, space.numels_t, 0.01, 0.9, 0.0)
+    torch.cuda.synchronize()
+    assert torch.isfinite(loss).item(), name
+    assert bool(torch.isfinite(space.grad).all().item()), name
