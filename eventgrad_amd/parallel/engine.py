@@ -268,12 +268,25 @@ class GossipEngine(CommEngine):
             fire = self.ctrl.step(norms, pass_num)
             if self.tracer is not None:
                 self.tracer.send_line(norms, self.ctrl.thres, fire)
-        mask = torch.from_numpy(fire.astype(np.uint8))
-        mask_l, mask_r = self.transport.exchange_masks(mask)
         my_fired = [i for i in range(self.space.sz) if fire[i]]
-        self._fired_l = [i for i in range(self.space.sz)
-                         if int(mask_l[i])]  # left neighbor fired
-        self._fired_r = [i for i in range(self.space.sz) if int(mask_r[i])]
+        # During warmup (pass < initial_comm_passes) and in decent mode the
+        # fire decision is deterministically all-ones on EVERY rank, so the
+        # blocking mask pre-exchange is skipped — the only host-serialized
+        # comm in the pipeline disappears for those passes.
+        ctrl = self.ctrl
+        all_fire_static = (ctrl.always_fire
+                           or pass_num < ctrl.initial_comm_passes)
+        if all_fire_static:
+            assert fire.all()
+            self._fired_l = list(range(self.space.sz))
+            self._fired_r = list(range(self.space.sz))
+        else:
+            mask = torch.from_numpy(fire.astype(np.uint8))
+            mask_l, mask_r = self.transport.exchange_masks(mask)
+            self._fired_l = [i for i in range(self.space.sz)
+                             if int(mask_l[i])]  # left neighbor fired
+            self._fired_r = [i for i in range(self.space.sz)
+                             if int(mask_r[i])]
         send_l, send_r = self._make_send(my_fired)
         self._recv_l = self._make_recv(self._fired_l)
         self._recv_r = self._make_recv(self._fired_r)
