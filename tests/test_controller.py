@@ -74,3 +74,13 @@ def test_state_roundtrip():
     m2 = c2.step(np.ones(3, np.float32), 8)
     assert (m1 == m2).all()
     assert np.array_equal(c.thres, c2.thres)
+
+
+def test_adaptive_horizon_zero_always_fires():
+    """dmnist/event/README.md:59-60: horizon 0 (adaptive) == the plain
+    Lian et al. ring — the pre-update thres*0 wipes the threshold every
+    pass even though firing writes a nonzero slope average."""
+    c = make(adaptive=True, horizon=0.0, warmup=0)
+    for p in range(1, 12):
+        fire = c.step(np.random.rand(3).astype(np.float32) * 10, p)
+        assert fire.all(), p
