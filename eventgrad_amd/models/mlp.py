@@ -24,4 +24,8 @@ class MLP(nn.Module):
             x = x.to(torch.bfloat16)
         x = O.relu(self.fc1(x))
         x = O.relu(self.fc2(x))
-        return x
+        # eval-mode log-prob head: the reference's test loop applies
+        # nll_loss(log_softmax(prediction)) itself (cent.cpp:192); our
+        # evaluate() expects log-probs from eval-mode forward, like
+        # cnn/resnet.
+        return x if self.training else O.log_softmax(x)

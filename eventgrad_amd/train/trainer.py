@@ -20,7 +20,8 @@ from ..data.sampler import (DistributedRandomSampler,
 from ..models import build_model
 from ..ops import functional as O
 from ..parallel import FlatParamSpace, build_engine, init_distributed
-from .checkpoint import load_checkpoint, save_checkpoint
+from .checkpoint import (load_checkpoint, rank_checkpoint_path,
+                         save_checkpoint)
 from .graphstep import FwdBwdGraph, can_graph
 from .metrics import RunMetrics
 from .trace import Tracer
@@ -59,9 +60,13 @@ class Trainer:
                                   num_tensors=self.space.sz)
         self.pass_num = 0
         self.start_epoch = 1
-        if cfg.resume and cfg.checkpoint_path:
-            st = load_checkpoint(cfg.checkpoint_path, self.space, self.engine,
-                                 self.device)
+        # per-rank checkpoint file: each rank's pre-consensus state differs
+        self.ckpt_path = (rank_checkpoint_path(cfg.checkpoint_path,
+                                               self.rank, self.world)
+                          if cfg.checkpoint_path else None)
+        if cfg.resume and self.ckpt_path:
+            st = load_checkpoint(self.ckpt_path, self.space, self.engine,
+                                 self.device, model=self.model)
             self.pass_num = st["pass_num"]
             self.start_epoch = st["epoch"] + 1
 
@@ -106,10 +111,11 @@ class Trainer:
                 print(f"{epoch}, {acc}", flush=True)  # decent.cpp:255 format
             if self.tracer:
                 self.tracer.train_line(epoch, acc, last_loss)
-            if (cfg.checkpoint_path and cfg.checkpoint_every_epochs
+            if (self.ckpt_path and cfg.checkpoint_every_epochs
                     and epoch % cfg.checkpoint_every_epochs == 0):
-                save_checkpoint(cfg.checkpoint_path, cfg, epoch,
-                                self.pass_num, self.space, self.engine)
+                save_checkpoint(self.ckpt_path, cfg, epoch,
+                                self.pass_num, self.space, self.engine,
+                                model=self.model)
         if self.device.type == "cuda":
             torch.cuda.synchronize()
         self.metrics.train_time_s = time.perf_counter() - t0
@@ -126,9 +132,10 @@ class Trainer:
             print(f"Total number of events - "
                   f"{self.metrics.num_events_total}", flush=True)
 
-        if cfg.checkpoint_path:
-            save_checkpoint(cfg.checkpoint_path, cfg, cfg.epochs,
-                            self.pass_num, self.space, self.engine)
+        if self.ckpt_path:
+            save_checkpoint(self.ckpt_path, cfg, cfg.epochs,
+                            self.pass_num, self.space, self.engine,
+                            model=self.model)
         if cfg.eval_at_end and self.rank == 0:
             self.evaluate()
         if self.tracer:
