@@ -46,6 +46,9 @@ def main() -> int:
     ap.add_argument("--backend", default=None,
                     help="dist backend override (testing: gloo lets two "
                          "ranks share one GPU; default nccl=RCCL)")
+    ap.add_argument("--dtype", default="bf16", choices=["bf16", "fp32"],
+                    help="GPU compute path (fp32 = torch/MIOpen "
+                         "full-precision; used for parity runs)")
     args = ap.parse_args()
 
     cfg = preset("dcifar10-event")
@@ -53,6 +56,8 @@ def main() -> int:
     cfg.model = args.model
     cfg.data.global_batch = None
     cfg.data.batch_size = args.batch
+    cfg.compute_dtype = args.dtype
+    O.set_compute_dtype(args.dtype)
 
     rank, world, device = init_distributed("auto", backend=args.backend)
     if device.type != "cuda":
@@ -76,7 +81,7 @@ def main() -> int:
 
     pass_num = 0
     graph = None
-    if not args.no_graph:
+    if not args.no_graph and args.dtype == "bf16":
         from eventgrad_amd.train.graphstep import FwdBwdGraph, can_graph
         if can_graph(model, device):
             graph = FwdBwdGraph(model, space, tuple(xs[0].shape), device)
@@ -133,7 +138,7 @@ def main() -> int:
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,
-            "dtype": "bf16",
+            "dtype": args.dtype,
             "data": "synthetic",
             "config": {
                 "model": ("resnet18-quirk (ref ResNet<BasicBlock>{2,2,2,2}, "

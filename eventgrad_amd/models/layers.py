@@ -42,7 +42,7 @@ class Conv2d(nn.Module):
         # fused-BN-stats epilogue requires out_ch % 64 == 0 (csrc/conv.hip);
         # the paired BatchNorm downgrades stats_ready by the same condition
         return O.conv2d(x, self.weight, self.bias, self.stride, self.padding,
-                        bn_stats=bn_stats and x.is_cuda and self.training
+                        bn_stats=bn_stats and O.use_native(x) and self.training
                         and self.out_ch % 64 == 0)
 
     def extra_repr(self):
@@ -71,13 +71,20 @@ class BatchNorm2d(nn.Module):
         super()._load_from_state_dict(*args, **kwargs)
         self._nbt = int(self.num_batches_tracked.item())
 
+    # checkpoint.py hooks (named_buffers() bypasses the state-dict hooks)
+    def sync_buffers_for_save(self):
+        self.num_batches_tracked.fill_(self._nbt)
+
+    def sync_buffers_after_load(self):
+        self._nbt = int(self.num_batches_tracked.item())
+
     def forward(self, x, fuse_relu: bool = False, stats_ready: bool = False):
         if self.training:
             self._nbt += 1
         return O.batch_norm(x, self.weight, self.bias, self.running_mean,
                             self.running_var, self.training, self.momentum,
                             self.eps, fuse_relu,
-                            stats_ready and x.is_cuda and self.training
+                            stats_ready and O.use_native(x) and self.training
                             and self.num_features % 64 == 0)
 
 

@@ -20,7 +20,7 @@ class MLP(nn.Module):
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         x = x.reshape(x.shape[0], 784)
-        if x.is_cuda:
+        if O.use_native(x):
             x = x.to(torch.bfloat16)
         x = O.relu(self.fc1(x))
         x = O.relu(self.fc2(x))

@@ -44,6 +44,9 @@ def save_checkpoint(path: str, cfg, epoch: int, pass_num: int, space, engine,
     }
     if model is not None:
         # non-parameter state: BN running stats + num_batches_tracked
+        for m in model.modules():
+            if hasattr(m, "sync_buffers_for_save"):
+                m.sync_buffers_for_save()
         state["buffers"] = {name: buf.detach().cpu().clone()
                             for name, buf in model.named_buffers()}
     if torch.cuda.is_available() and space.param.is_cuda:
@@ -76,6 +79,9 @@ def load_checkpoint(path: str, space, engine, device, model=None) -> dict:
         for name, saved in state["buffers"].items():
             if name in bufs:
                 bufs[name].copy_(saved.to(bufs[name].device))
+        for m in model.modules():
+            if hasattr(m, "sync_buffers_after_load"):
+                m.sync_buffers_after_load()
     ed = state.get("engine", {})
     ctrl = getattr(engine, "ctrl", None)
     if ctrl is not None and "controller" in ed:

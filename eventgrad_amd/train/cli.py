@@ -51,6 +51,9 @@ def build_parser() -> argparse.ArgumentParser:
     p.add_argument("--checkpoint-every", type=int, default=None)
     p.add_argument("--resume", action="store_true", default=None)
     p.add_argument("--device", default=None)
+    p.add_argument("--compute-dtype", choices=["bf16", "fp32"], default=None,
+                   help="GPU compute path: bf16 = native HIP kernels "
+                        "(default), fp32 = full-precision torch/MIOpen")
     p.add_argument("--no-eval", action="store_true", default=None)
     p.add_argument("--json-out", default=None,
                    help="write metrics summary JSON here (rank 0)")
@@ -77,6 +80,8 @@ def config_from_args(args) -> RunConfig:
         cfg.trace = True
     if args.no_eval:
         cfg.eval_at_end = False
+    if args.compute_dtype is not None:
+        cfg.compute_dtype = args.compute_dtype
     if args.dataset is not None:
         cfg.data.dataset = args.dataset
     if args.data_path is not None:

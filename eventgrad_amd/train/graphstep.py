@@ -31,6 +31,10 @@ def can_graph(model, device) -> bool:
         return False
     if getattr(model, "uses_dropout", False):
         return False
+    # fp32 mode runs aten/MIOpen ops; capture of library convs (workspace
+    # allocation, autotuning) is not graph-safe — eager only.
+    if O.get_compute_dtype() != "bf16":
+        return False
     return True
 
 

@@ -32,6 +32,7 @@ class Trainer:
         self.cfg = cfg
         self.rank, self.world, self.device = init_distributed(
             cfg.device, backend=cfg.dist_backend)
+        O.set_compute_dtype(cfg.compute_dtype)
         torch.manual_seed(cfg.seed)  # ref: torch::manual_seed(0) everywhere
 
         self.model = build_model(cfg.model, cfg.data.num_classes)

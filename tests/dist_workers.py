@@ -18,7 +18,7 @@ def _small_cfg(mode: str, **kw):
     cfg = RunConfig(
         mode=mode, model=kw.pop("model", "cnn2"),
         epochs=kw.pop("epochs", 2), device="cpu", trigger=trig,
-        data=DataConfig(dataset="synthetic-mnist",
+        data=DataConfig(dataset=kw.pop("dataset", "synthetic-mnist"),
                         batch_size=kw.pop("batch_size", 32),
                         synthetic_train_samples=kw.pop("n_train", 256),
                         synthetic_test_samples=64),
@@ -127,3 +127,51 @@ def checkpoint_resume_worker(rank, world, port, outdir):
 
     torch.save({"ref": ref, "resumed": tr2.space.param.clone()},
                os.path.join(outdir, f"ckres_r{rank}.pt"))
+
+
+def checkpoint_resume_bn_worker(rank, world, port, outdir):
+    """Resume identity for a BatchNorm model (resnet20): params AND the BN
+    running stats (registered buffers, outside FlatParamSpace) must round-trip
+    through the checkpoint, so post-resume eval matches the straight run."""
+    init_env(rank, world, port)
+    from eventgrad_amd.train.trainer import Trainer
+
+    kw = dict(model="resnet20", dataset="synthetic", batch_size=16,
+              n_train=64, lr=0.05, momentum=0.9)
+    ck = os.path.join(outdir, "ckbn.pt")  # single path; Trainer adds .rank{r}
+
+    cfg = _small_cfg("event", epochs=2, **kw)
+    tr = Trainer(cfg)
+    tr.train()
+    ref_param = tr.space.param.clone()
+    ref_bufs = {k: v.detach().clone() for k, v in tr.model.named_buffers()}
+    ref_eval = tr.evaluate() if rank == 0 else None
+    ref_loss = tr.metrics.test_loss if rank == 0 else None
+    torch.distributed.destroy_process_group()
+
+    os.environ["MASTER_PORT"] = str(port + 1)
+    cfg1 = _small_cfg("event", epochs=1, **kw)
+    cfg1.checkpoint_path = ck
+    cfg1.final_consensus = False
+    tr1 = Trainer(cfg1)
+    tr1.train()
+    torch.distributed.destroy_process_group()
+
+    os.environ["MASTER_PORT"] = str(port + 2)
+    cfg2 = _small_cfg("event", epochs=2, **kw)
+    cfg2.checkpoint_path = ck
+    cfg2.resume = True
+    tr2 = Trainer(cfg2)
+    tr2.train()
+    res_bufs = dict(tr2.model.named_buffers())
+    bufs_equal = all(torch.equal(ref_bufs[k], res_bufs[k].detach())
+                     for k in ref_bufs)
+    res_eval = tr2.evaluate() if rank == 0 else None
+    res_loss = tr2.metrics.test_loss if rank == 0 else None
+    torch.distributed.destroy_process_group()
+
+    torch.save({"ref": ref_param, "resumed": tr2.space.param.clone(),
+                "bufs_equal": bufs_equal, "n_bufs": len(ref_bufs),
+                "ref_eval": ref_eval, "res_eval": res_eval,
+                "ref_loss": ref_loss, "res_loss": res_loss},
+               os.path.join(outdir, f"ckbn_r{rank}.pt"))

@@ -109,3 +109,33 @@ def test_cifar_binary_parser(tmp_path):
     assert len(ds) == 25
     x, y = ds[0]
     assert x.shape == (3, 32, 32) and 0 <= y < 10
+
+
+def test_compute_dtype_plumbing():
+    """RunConfig.compute_dtype is consumed: it toggles the ops dispatch
+    (the flag must not be dead)."""
+    import pytest
+    import torch
+    from eventgrad_amd.ops import functional as O
+
+    assert O.get_compute_dtype() == "bf16"
+    O.set_compute_dtype("fp32")
+    try:
+        assert O.get_compute_dtype() == "fp32"
+        # CPU tensors never take the native path in either mode
+        assert not O.use_native(torch.zeros(2))
+        from eventgrad_amd.train.graphstep import can_graph
+
+        class _M:
+            uses_dropout = False
+        assert not can_graph(_M(), torch.device("cpu"))
+        with pytest.raises(ValueError):
+            O.set_compute_dtype("fp16")
+    finally:
+        O.set_compute_dtype("bf16")
+
+
+def test_cli_compute_dtype_flag():
+    from eventgrad_amd.train.cli import build_parser, config_from_args
+    args = build_parser().parse_args(["--compute-dtype", "fp32"])
+    assert config_from_args(args).compute_dtype == "fp32"

@@ -54,3 +54,18 @@ def test_checkpoint_resume_identity(tmp_path):
         d = torch.load(f, weights_only=False)
         assert torch.allclose(d["ref"], d["resumed"], atol=1e-6), \
             (d["ref"] - d["resumed"]).abs().max()
+
+
+def test_checkpoint_resume_bn_model(tmp_path):
+    """BN running stats (buffers outside FlatParamSpace) must survive the
+    checkpoint round-trip: params bit-identical AND eval identical."""
+    run_world(W.checkpoint_resume_bn_worker, 2, str(tmp_path))
+    files = sorted(glob.glob(os.path.join(tmp_path, "ckbn_r*.pt")))
+    assert len(files) == 2
+    for f in files:
+        d = torch.load(f, weights_only=False)
+        assert torch.allclose(d["ref"], d["resumed"], atol=1e-6)
+        assert d["n_bufs"] > 0 and d["bufs_equal"]
+        if d["ref_eval"] is not None:  # rank 0 evaluated
+            assert d["res_eval"] == d["ref_eval"]
+            assert abs(d["res_loss"] - d["ref_loss"]) < 1e-6
