@@ -140,11 +140,18 @@ def checkpoint_resume_bn_worker(rank, world, port, outdir):
               n_train=64, lr=0.05, momentum=0.9)
     ck = os.path.join(outdir, "ckbn.pt")  # single path; Trainer adds .rank{r}
 
+    def _bufs(model):
+        # sync python-side mirrors (num_batches_tracked) before harvesting
+        for m in model.modules():
+            if hasattr(m, "sync_buffers_for_save"):
+                m.sync_buffers_for_save()
+        return {k: v.detach().clone() for k, v in model.named_buffers()}
+
     cfg = _small_cfg("event", epochs=2, **kw)
     tr = Trainer(cfg)
     tr.train()
     ref_param = tr.space.param.clone()
-    ref_bufs = {k: v.detach().clone() for k, v in tr.model.named_buffers()}
+    ref_bufs = _bufs(tr.model)
     ref_eval = tr.evaluate() if rank == 0 else None
     ref_loss = tr.metrics.test_loss if rank == 0 else None
     torch.distributed.destroy_process_group()
@@ -163,8 +170,8 @@ def checkpoint_resume_bn_worker(rank, world, port, outdir):
     cfg2.resume = True
     tr2 = Trainer(cfg2)
     tr2.train()
-    res_bufs = dict(tr2.model.named_buffers())
-    bufs_equal = all(torch.equal(ref_bufs[k], res_bufs[k].detach())
+    res_bufs = _bufs(tr2.model)
+    bufs_equal = all(torch.equal(ref_bufs[k], res_bufs[k])
                      for k in ref_bufs)
     res_eval = tr2.evaluate() if rank == 0 else None
     res_loss = tr2.metrics.test_loss if rank == 0 else None
