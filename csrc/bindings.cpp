@@ -17,6 +17,8 @@ void scatter_segments(torch::Tensor, torch::Tensor, torch::Tensor,
 torch::Tensor trigger_update(torch::Tensor, torch::Tensor, torch::Tensor,
                              torch::Tensor, torch::Tensor, torch::Tensor,
                              long, bool, double, double, long, bool);
+torch::Tensor trigger_decide(torch::Tensor, torch::Tensor, torch::Tensor,
+                             long, bool, double, double, long);
 // elementwise.hip
 torch::Tensor relu_fwd(torch::Tensor);
 torch::Tensor relu_bwd(torch::Tensor, torch::Tensor);
@@ -69,6 +71,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gather_segments", &eg::gather_segments);
   m.def("scatter_segments", &eg::scatter_segments);
   m.def("trigger_update", &eg::trigger_update);
+  m.def("trigger_decide", &eg::trigger_decide);
   m.def("relu_fwd", &eg::relu_fwd);
   m.def("relu_bwd", &eg::relu_bwd);
   m.def("add_relu_fwd", &eg::add_relu_fwd);

@@ -269,6 +269,38 @@ __global__ void trigger_update_kernel(
   mask[i] = fire ? 1 : 0;
 }
 
+// Pure fire decision — no state mutation. Bit-identical to the decision
+// trigger_update takes (same float expressions), so it can run one pass
+// EARLY (at the end of the previous optimizer step) to post the mask
+// exchange off the host critical path; trigger_update commits later.
+__global__ void trigger_decide_kernel(
+    const float* __restrict__ norms_sq, const float* __restrict__ thres,
+    const float* __restrict__ last_sent_norm, unsigned char* __restrict__ mask,
+    int sz, float pass_num, int adaptive, float horizon, float constant,
+    int warmup) {
+  int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= sz) return;
+  float norm = sqrtf(norms_sq[i]);
+  float value_diff = fabsf(norm - last_sent_norm[i]);
+  float th = adaptive ? thres[i] * horizon : constant;
+  mask[i] = ((value_diff >= th) || (pass_num < (float)warmup)) ? 1 : 0;
+}
+
+torch::Tensor trigger_decide(torch::Tensor norms_sq, torch::Tensor thres,
+                             torch::Tensor last_sent_norm, long pass_num,
+                             bool adaptive, double horizon, double constant,
+                             long warmup) {
+  CHECK_IN(norms_sq);
+  int sz = (int)norms_sq.numel();
+  auto mask = torch::empty({sz}, norms_sq.options().dtype(torch::kUInt8));
+  trigger_decide_kernel<<<ceil_div(sz, 128), 128, 0, cur_stream()>>>(
+      norms_sq.data_ptr<float>(), thres.data_ptr<float>(),
+      last_sent_norm.data_ptr<float>(), mask.data_ptr<unsigned char>(), sz,
+      (float)pass_num, adaptive ? 1 : 0, (float)horizon, (float)constant,
+      (int)warmup);
+  return mask;
+}
+
 torch::Tensor trigger_update(torch::Tensor norms_sq, torch::Tensor thres,
                              torch::Tensor last_sent_norm,
                              torch::Tensor last_sent_iter,

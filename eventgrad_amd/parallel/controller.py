@@ -50,6 +50,23 @@ class TriggerController:
         self.last_norms = np.zeros(sz, dtype=np.float32)
         self.last_fired = np.zeros(sz, dtype=bool)
 
+    def decide(self, norms: np.ndarray, pass_num: int) -> np.ndarray:
+        """Pure fire decision for pass_num — NO state mutation.
+
+        Computes exactly the mask step() would produce from the current
+        state (same float ops), so the decision can be taken early (at the
+        end of the previous optimizer step) to post the mask exchange
+        off the critical path; step() later commits the state change.
+        """
+        if self.always_fire:
+            return np.ones(self.sz, dtype=bool)
+        norms = norms.astype(np.float32, copy=False)
+        value_diff = np.abs(norms - self.last_sent_norm)
+        thres_eff = (self.thres * self.horizon if self.adaptive
+                     else np.full(self.sz, self.constant, dtype=np.float32))
+        return (value_diff >= thres_eff) | (pass_num <
+                                            self.initial_comm_passes)
+
     def step(self, norms: np.ndarray, pass_num: int) -> np.ndarray:
         """Evaluate the trigger for one pass; returns fire mask (bool[sz])."""
         norms = norms.astype(np.float32, copy=False)
@@ -141,8 +158,24 @@ class GpuTriggerController:
     def num_events(self, v: int):
         self._num_events.fill_(int(v))
 
-    def step_device(self, norms_sq, pass_num: int):
-        """norms_sq: device fp32[sz] (squared L2 norms). Returns np bool[sz]."""
+    def decide_device(self, norms_sq, pass_num: int):
+        """Pure fire decision (no state mutation) — the device counterpart
+        of TriggerController.decide. Returns np bool[sz] (one small D2H)."""
+        import numpy as _np
+
+        if self.always_fire:
+            return _np.ones(self.sz, dtype=bool)
+        from ..ops.backend import native
+
+        mask = native().trigger_decide(
+            norms_sq, self.thres, self.last_sent_norm, pass_num,
+            self.adaptive, self.horizon, self.constant,
+            self.initial_comm_passes)
+        return mask.cpu().numpy().astype(bool)
+
+    def step_device(self, norms_sq, pass_num: int, need_mask: bool = True):
+        """norms_sq: device fp32[sz] (squared L2 norms). Returns np bool[sz]
+        (or None when need_mask=False — commit-only, no D2H sync)."""
         from ..ops.backend import native
 
         mask = native().trigger_update(
@@ -151,6 +184,8 @@ class GpuTriggerController:
             self.horizon, self.constant, self.initial_comm_passes,
             self.always_fire)
         self.last_norms = norms_sq
+        if not need_mask:
+            return None
         fired = mask.cpu().numpy().astype(bool)
         self.last_fired = fired
         return fired

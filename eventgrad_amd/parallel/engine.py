@@ -236,6 +236,9 @@ class GossipEngine(CommEngine):
         self._recv_r = None
         self._fired_l: list = []
         self._fired_r: list = []
+        # lookahead: (pass_num, fire mask) decided at the end of the
+        # previous step(), whose mask exchange is already in flight
+        self._pending: tuple | None = None
 
     # -- send-side payload construction (dense) --------------------------
     def _make_send(self, fired_idx):
@@ -250,39 +253,78 @@ class GossipEngine(CommEngine):
     def _apply_recv(self, payload, fired_idx, inbox):
         self.K.unpack(self.space, payload, fired_idx, inbox)
 
+    def _sq(self):
+        sq = getattr(self, "_last_norms_sq", None)
+        if sq is None:
+            sq = self.K.sqnorms(self.space, self.space.param)
+        return sq
+
+    def _all_fire_static(self, pass_num: int) -> bool:
+        """True when the fire decision is deterministically all-ones on
+        EVERY rank (decent mode / warmup) — no mask exchange needed."""
+        return (self.ctrl.always_fire
+                or pass_num < self.ctrl.initial_comm_passes)
+
+    def step(self) -> None:
+        super().step()
+        # lookahead: decide pass+1's trigger NOW (pure, no state commit)
+        # and post the tiny mask exchange so its wire time overlaps the
+        # host work between passes (accuracy sync, batch prep) instead of
+        # blocking at the top of begin_pass. The commit happens in
+        # begin_pass; a decision whose pass never runs commits nothing.
+        if self.world <= 1:
+            return
+        nxt = self.pass_num + 1
+        if self._all_fire_static(nxt):
+            return  # begin_pass takes the static fast path, no exchange
+        if self.device.type == "cuda":
+            fire = self.ctrl.decide_device(self._last_norms_sq, nxt)
+        else:
+            fire = self.ctrl.decide(self._norms(), nxt)
+        self.transport.post_masks(torch.from_numpy(fire.astype(np.uint8)))
+        self._pending = (nxt, fire)
+
     def begin_pass(self, pass_num: int) -> None:
         self.pass_num = pass_num
         if self.world <= 1:
             return
-        if self.device.type == "cuda":
+        pending = self._pending
+        self._pending = None
+        have_lookahead = pending is not None and pending[0] == pass_num
+        if have_lookahead:
+            # commit the controller state for the pre-decided mask; the
+            # commit recomputes the identical decision from the same
+            # (unchanged) state + norms, so no second mask D2H is needed
+            fire = pending[1]
+            if self.device.type == "cuda":
+                self.ctrl.step_device(self._sq(), pass_num, need_mask=False)
+                self.ctrl.last_fired = fire
+            else:
+                committed = self.ctrl.step(self._norms(), pass_num)
+                assert (committed == fire).all()
+        elif self.device.type == "cuda":
             # device-resident controller: only the mask crosses to host
-            sq = getattr(self, "_last_norms_sq", None)
-            if sq is None:
-                sq = self.K.sqnorms(self.space, self.space.param)
-            fire = self.ctrl.step_device(sq, pass_num)
-            if self.tracer is not None:
-                norms, thres = self.ctrl.trace_values()
-                self.tracer.send_line(norms, thres, fire)
+            fire = self.ctrl.step_device(self._sq(), pass_num)
         else:
-            norms = self._norms()
-            fire = self.ctrl.step(norms, pass_num)
-            if self.tracer is not None:
-                self.tracer.send_line(norms, self.ctrl.thres, fire)
+            fire = self.ctrl.step(self._norms(), pass_num)
+        if self.tracer is not None:
+            if self.device.type == "cuda":
+                norms, thres = self.ctrl.trace_values()
+            else:
+                norms, thres = self.ctrl.last_norms, self.ctrl.thres
+            self.tracer.send_line(norms, thres, fire)
         my_fired = [i for i in range(self.space.sz) if fire[i]]
-        # During warmup (pass < initial_comm_passes) and in decent mode the
-        # fire decision is deterministically all-ones on EVERY rank, so the
-        # blocking mask pre-exchange is skipped — the only host-serialized
-        # comm in the pipeline disappears for those passes.
-        ctrl = self.ctrl
-        all_fire_static = (ctrl.always_fire
-                           or pass_num < ctrl.initial_comm_passes)
-        if all_fire_static:
+        if self._all_fire_static(pass_num):
+            # all-ones on every rank: skip the mask exchange entirely
             assert fire.all()
             self._fired_l = list(range(self.space.sz))
             self._fired_r = list(range(self.space.sz))
         else:
-            mask = torch.from_numpy(fire.astype(np.uint8))
-            mask_l, mask_r = self.transport.exchange_masks(mask)
+            if have_lookahead:
+                mask_l, mask_r = self.transport.wait_masks()
+            else:  # first pass / post-resume fallback: blocking exchange
+                mask = torch.from_numpy(fire.astype(np.uint8))
+                mask_l, mask_r = self.transport.exchange_masks(mask)
             self._fired_l = [i for i in range(self.space.sz)
                              if int(mask_l[i])]  # left neighbor fired
             self._fired_r = [i for i in range(self.space.sz)
@@ -302,6 +344,14 @@ class GossipEngine(CommEngine):
         if self.tracer is not None:
             self._trace_recv()
         self.K.avg3(self.space.param, self.inbox_left, self.inbox_right)
+
+    def finalize(self) -> dict:
+        # drain a lookahead mask exchange whose pass never ran (training
+        # ended): peers posted symmetrically, so waiting completes it
+        if self.transport is not None:
+            self.transport.cancel_pending_masks()
+        self._pending = None
+        return super().finalize()
 
     def _trace_recv(self):
         # reference logs, per tensor: new-msg flag + received-half norm

@@ -50,6 +50,7 @@ class RingTransport:
         self.use_tags = self.backend == "gloo"
         self._reqs: List = []
         self._recv_copies: List = []  # (host_staging, device_dst) for gloo+GPU
+        self._mask_state = None       # in-flight post_masks state
 
     def _ops(self, pairs):
         """pairs: list of (kind, tensor, peer, tag). Returns work handles.
@@ -76,10 +77,16 @@ class RingTransport:
         return dist.batch_isend_irecv(ops)
 
     # ------------------------------------------------------------------
-    def exchange_masks(self, mask: torch.Tensor):
-        """Blocking exchange of the fire mask (uint8[sz]) with both neighbors.
+    # mask exchange: split post/wait so the tiny phase-A transfer can be
+    # posted at the END of the previous pass's optimizer step and overlap
+    # the host-side work between passes (accuracy sync, next-batch prep) —
+    # "move the per-pass blocking mask exchange off the host
+    # critical path".
+    def post_masks(self, mask: torch.Tensor) -> None:
+        """Post the nonblocking mask exchange (uint8[sz]) with both neighbors.
 
-        Returns (mask_from_left, mask_from_right) as uint8 CPU tensors.
+        Must be matched by wait_masks() before the next post. The mask is
+        kept referenced until wait_masks so the send buffer stays alive.
         """
         m = mask.to(self.device, dtype=torch.uint8).contiguous()
         fr_right = torch.empty_like(m)
@@ -90,12 +97,36 @@ class RingTransport:
             ("recv", fr_right, self.right, TAG_MASK + TAG_L),
             ("recv", fr_left, self.left, TAG_MASK + TAG_R),
         ]
-        for r in self._ops(pairs):
+        # keep any in-flight payload staging pairs separate from the mask's
+        saved, self._recv_copies = self._recv_copies, []
+        reqs = self._ops(pairs)
+        self._mask_state = (reqs, m, fr_left, fr_right, self._recv_copies)
+        self._recv_copies = saved
+
+    def wait_masks(self):
+        """Complete a posted mask exchange.
+
+        Returns (mask_from_left, mask_from_right) as uint8 CPU tensors.
+        """
+        reqs, _m, fr_left, fr_right, copies = self._mask_state
+        self._mask_state = None
+        for r in reqs:
             r.wait()
-        for host, dev in self._recv_copies:
+        for host, dev in copies:
             dev.copy_(host)
-        self._recv_copies = []
         return fr_left.cpu(), fr_right.cpu()
+
+    def cancel_pending_masks(self) -> None:
+        """Drain a posted mask exchange whose pass never ran (end of
+        training): the peers posted symmetrically, so the wire completes —
+        just wait and discard before the closing collectives."""
+        if getattr(self, "_mask_state", None) is not None:
+            self.wait_masks()
+
+    def exchange_masks(self, mask: torch.Tensor):
+        """Blocking mask exchange (fallback when no lookahead was posted)."""
+        self.post_masks(mask)
+        return self.wait_masks()
 
     def post_payloads(self, send_l: Optional[torch.Tensor],
                       send_r: Optional[torch.Tensor],

@@ -72,6 +72,26 @@ def event_equals_decent_worker(rank, world, port, outdir):
                os.path.join(outdir, f"eqdec_r{rank}.pt"))
 
 
+def event_twice_deterministic_worker(rank, world, port, outdir):
+    """Run the same post-warmup sparse-mask event config twice; the matched
+    mask+payload protocol must be fully deterministic (identical params and
+    identical event counts), at any world size."""
+    init_env(rank, world, port)
+    from eventgrad_amd.train.trainer import Trainer
+
+    results = []
+    for cycle in range(2):
+        os.environ["MASTER_PORT"] = str(port + cycle)
+        cfg = _small_cfg("event", epochs=2, momentum=0.9)
+        tr = Trainer(cfg)
+        m = tr.train()
+        results.append((tr.space.param.clone(), m.num_events_total))
+        torch.distributed.destroy_process_group()
+    torch.save({"identical": torch.equal(results[0][0], results[1][0]),
+                "events0": results[0][1], "events1": results[1][1]},
+               os.path.join(outdir, f"det_r{rank}.pt"))
+
+
 def cent_equals_fullbatch_worker(rank, world, port, outdir):
     """cent (allreduce-averaged grads over equal shards) must equal a serial
     run on the concatenated data with the same lr (cent.cpp:130-145
