@@ -84,3 +84,20 @@ def test_adaptive_horizon_zero_always_fires():
     for p in range(1, 12):
         fire = c.step(np.random.rand(3).astype(np.float32) * 10, p)
         assert fire.all(), p
+
+
+def test_decide_matches_step_mask():
+    """decide() (pure lookahead) must equal the mask step() then commits,
+    from any reachable state."""
+    import numpy as np
+    from eventgrad_amd.parallel.controller import TriggerController
+
+    rng = np.random.default_rng(3)
+    for adaptive in (True, False):
+        ctrl = TriggerController(6, adaptive=adaptive, horizon=1.03,
+                                 constant=5e-4, initial_comm_passes=4)
+        for p in range(1, 50):
+            norms = rng.random(6).astype(np.float32) * (1 + p / 20)
+            d = ctrl.decide(norms, p)
+            f = ctrl.step(norms, p)
+            assert (d == f).all(), (adaptive, p)

@@ -19,10 +19,22 @@ class LoopbackTransport:
 
     def __init__(self):
         self._sent = None
+        self._mask = None
 
     def exchange_masks(self, mask):
         m = mask.cpu()
         return m.clone(), m.clone()
+
+    # lookahead protocol (engine.step posts next pass's mask)
+    def post_masks(self, mask):
+        self._mask = mask.cpu()
+
+    def wait_masks(self):
+        m, self._mask = self._mask, None
+        return m.clone(), m.clone()
+
+    def cancel_pending_masks(self):
+        self._mask = None
 
     def post_payloads(self, send_l, send_r, recv_l, recv_r):
         if send_l is not None and recv_l is not None and send_l.numel():

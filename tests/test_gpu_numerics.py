@@ -321,6 +321,25 @@ def test_trigger_update_matches_host_controller():
     assert int(nev.item()) == host.num_events
 
 
+def test_trigger_decide_matches_update_mask():
+    """The pure decide kernel (lookahead mask posting) must produce exactly
+    the mask trigger_update commits from the same state."""
+    sz, H = 8, 2
+    thres = torch.zeros(sz, device=DEV)
+    lsn = torch.zeros(sz, device=DEV)
+    lsi = torch.zeros(sz, device=DEV)
+    slopes = torch.zeros(sz * H, device=DEV)
+    nev = torch.zeros(1, dtype=torch.int32, device=DEV)
+    rng = np.random.default_rng(7)
+    for p in range(1, 40):
+        nd = torch.tensor((rng.random(sz).astype(np.float32)
+                           * (1 + p / 10)) ** 2, device=DEV)
+        md = core().trigger_decide(nd, thres, lsn, p, True, 1.05, 5e-4, 3)
+        mu = core().trigger_update(nd, thres, lsn, lsi, slopes, nev, p, True,
+                                   1.05, 5e-4, 3, False)
+        assert torch.equal(md, mu), p
+
+
 def test_topk_absdiff():
     torch.manual_seed(11)
     n, k = 10000, 77
