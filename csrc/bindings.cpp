@@ -26,9 +26,13 @@ torch::Tensor add_relu_fwd(torch::Tensor, torch::Tensor);
 std::vector<torch::Tensor> dropout_fwd(torch::Tensor, double, long, bool);
 torch::Tensor dropout_bwd(torch::Tensor, torch::Tensor, double, bool);
 torch::Tensor channel_sum(torch::Tensor);
+void channel_sum_into(torch::Tensor, torch::Tensor);
 torch::Tensor oihw_to_krsc(torch::Tensor);
 torch::Tensor krsc_to_crsk(torch::Tensor);
 torch::Tensor krsc_to_oihw(torch::Tensor);
+void refresh_conv_shadows(torch::Tensor, torch::Tensor, torch::Tensor,
+                          torch::Tensor, torch::Tensor, torch::Tensor,
+                          torch::Tensor, torch::Tensor, torch::Tensor);
 // gemm.hip
 torch::Tensor gemm_bias(torch::Tensor, torch::Tensor, torch::Tensor, bool);
 // conv.hip
@@ -38,13 +42,15 @@ torch::Tensor conv2d_dgrad(torch::Tensor, torch::Tensor, long, long, long,
                            long);
 torch::Tensor conv2d_wgrad(torch::Tensor, torch::Tensor, long, long, long,
                            long);
+void conv2d_wgrad_into(torch::Tensor, torch::Tensor, torch::Tensor, long,
+                       long, long, long);
 // bn.hip
 std::vector<torch::Tensor> bn_fwd(torch::Tensor, torch::Tensor, torch::Tensor,
                                   torch::Tensor, torch::Tensor, double,
                                   double, bool, bool, bool);
 std::vector<torch::Tensor> bn_bwd(torch::Tensor, torch::Tensor, torch::Tensor,
                                   torch::Tensor, torch::Tensor, torch::Tensor,
-                                  bool, bool);
+                                  bool, bool, torch::Tensor, torch::Tensor);
 // pool.hip
 std::vector<torch::Tensor> maxpool2x2_fwd(torch::Tensor);
 torch::Tensor maxpool2x2_bwd(torch::Tensor, torch::Tensor, long, long);
@@ -78,20 +84,26 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("dropout_fwd", &eg::dropout_fwd);
   m.def("dropout_bwd", &eg::dropout_bwd);
   m.def("channel_sum", &eg::channel_sum);
+  m.def("channel_sum_into", &eg::channel_sum_into);
   m.def("oihw_to_krsc", &eg::oihw_to_krsc);
   m.def("krsc_to_crsk", &eg::krsc_to_crsk);
   m.def("krsc_to_oihw", &eg::krsc_to_oihw);
+  m.def("refresh_conv_shadows", &eg::refresh_conv_shadows);
   m.def("gemm_bias", &eg::gemm_bias);
   m.def("conv2d_fwd", &eg::conv2d_fwd, py::arg("x"), py::arg("w"),
         py::arg("bias"), py::arg("stride"), py::arg("pad"),
         py::arg("collect_bn_stats") = false);
   m.def("conv2d_dgrad", &eg::conv2d_dgrad);
   m.def("conv2d_wgrad", &eg::conv2d_wgrad);
+  m.def("conv2d_wgrad_into", &eg::conv2d_wgrad_into);
   m.def("bn_fwd", &eg::bn_fwd, py::arg("x"), py::arg("gamma"), py::arg("beta"),
         py::arg("running_mean"), py::arg("running_var"), py::arg("momentum"),
         py::arg("eps"), py::arg("training"), py::arg("relu"),
         py::arg("have_stats") = false);
-  m.def("bn_bwd", &eg::bn_bwd);
+  m.def("bn_bwd", &eg::bn_bwd, py::arg("dy"), py::arg("x"), py::arg("mean"),
+        py::arg("invstd"), py::arg("gamma"), py::arg("y"), py::arg("relu"),
+        py::arg("training"), py::arg("dgamma_out") = torch::Tensor(),
+        py::arg("dbeta_out") = torch::Tensor());
   m.def("maxpool2x2_fwd", &eg::maxpool2x2_fwd);
   m.def("maxpool2x2_bwd", &eg::maxpool2x2_bwd);
   m.def("avgpool_fwd", &eg::avgpool_fwd);

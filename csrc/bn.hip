@@ -394,15 +394,29 @@ std::vector<torch::Tensor> bn_fwd(torch::Tensor x, torch::Tensor gamma,
 std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor x,
                                   torch::Tensor mean, torch::Tensor invstd,
                                   torch::Tensor gamma, torch::Tensor y,
-                                  bool relu, bool training) {
+                                  bool relu, bool training,
+                                  torch::Tensor dgamma_out,
+                                  torch::Tensor dbeta_out) {
   CHECK_IN(dy); CHECK_IN(x);
   int c = (int)x.size(-1);
   long rows = x.numel() / c;
   long n = x.numel();
   auto f32 = x.options().dtype(torch::kFloat32);
-  auto sums = torch::zeros({2 * c}, f32);
-  auto sum_dy = sums.narrow(0, 0, c);
-  auto sum_dyx = sums.narrow(0, c, c);
+  // dgamma == sum(dy'*xhat), dbeta == sum(dy'): when the flat-grad views
+  // are supplied (pre-zeroed by the step's zero_grad), the reduction
+  // accumulates the gradients in place — no allocation, no aten
+  // accumulate-add afterwards.
+  torch::Tensor sum_dy, sum_dyx;
+  const bool direct = dgamma_out.defined() && dgamma_out.numel() == c;
+  if (direct) {
+    CHECK_IN(dgamma_out); CHECK_IN(dbeta_out);
+    sum_dyx = dgamma_out;
+    sum_dy = dbeta_out;
+  } else {
+    auto sums = torch::zeros({2 * c}, f32);
+    sum_dy = sums.narrow(0, 0, c);
+    sum_dyx = sums.narrow(0, c, c);
+  }
   auto stream = cur_stream();
   if (c % 8 == 0 && c <= bn::BN_MAXC) {
     int rpb = 256 / (c / 8);

@@ -349,7 +349,13 @@ __device__ __forceinline__ bf16x8 tr16_frag(const __bf16* base, int np0,
 __device__ __forceinline__ int np_img(int np) { return np * 16 + (np >> 3) * 8; }
 constexpr int WIMG = WBK * 16 + (WBK / 8) * 8 + 16;
 
-template <bool ATOMIC>
+// OIHW: write dw straight into the fp32 OIHW flat-grad view (accumulate
+// semantics into the pre-zeroed grad buffer) instead of a fresh [K, RSC]
+// KRSC tensor — removes the krsc_to_oihw transform AND the autograd
+// accumulate-add from the per-step path. The rsc
+// column decomposes once per fragment column; the OIHW offset is then
+// kk * CRS + (c*R + r)*S + s.
+template <bool ATOMIC, bool OIHW = false>
 __global__ __launch_bounds__(256) void conv_wgrad_fast_kernel(
     const bf16* __restrict__ x, const bf16* __restrict__ dy,
     float* __restrict__ dw, Geom g, long NP, long npslice) {
@@ -482,21 +488,27 @@ __global__ __launch_bounds__(256) void conv_wgrad_fast_kernel(
     for (int fj = 0; fj < 2; ++fj) {
       long nn = n0 + wc * 32 + fj * 16 + cn;
       if (nn >= RED) continue;
+      long col = nn;
+      if (OIHW) {
+        const int c = (int)(nn % g.C);
+        const int rs = (int)(nn / g.C);
+        col = ((long)c * g.R + rs / g.S) * g.S + rs % g.S;
+      }
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         long kk = k0c + wr * 32 + fi * 16 + cm + r;
         if (kk >= g.K) continue;
         if (ATOMIC) {
-          atomicAdd(&dw[kk * RED + nn], acc[fi][fj][r]);
+          atomicAdd(&dw[kk * RED + col], acc[fi][fj][r]);
         } else {
-          dw[kk * RED + nn] = acc[fi][fj][r];
+          dw[kk * RED + col] = acc[fi][fj][r];
         }
       }
     }
   }
 }
 
-template <bool FAST>  // generic fallback (any C/K); WBK_GEN=32 transposed
+template <bool FAST, bool OIHW = false>  // generic fallback (any C/K)
 __global__ __launch_bounds__(256) void conv_wgrad_kernel(
     const bf16* __restrict__ x, const bf16* __restrict__ dy,
     float* __restrict__ dw, Geom g, long NP, long npslice) {
@@ -632,11 +644,17 @@ __global__ __launch_bounds__(256) void conv_wgrad_kernel(
     for (int fj = 0; fj < 2; ++fj) {
       long nn = n0 + wc * 32 + fj * 16 + cn;
       if (nn >= RED) continue;
+      long col = nn;
+      if (OIHW) {
+        const int c = (int)(nn % g.C);
+        const int rs = (int)(nn / g.C);
+        col = ((long)c * g.R + rs / g.S) * g.S + rs % g.S;
+      }
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         long kk = k0c + wr * 32 + fi * 16 + cm + r;
         if (kk >= g.K) continue;
-        atomicAdd(&dw[kk * RED + nn], acc[fi][fj][r]);
+        atomicAdd(&dw[kk * RED + col], acc[fi][fj][r]);
       }
     }
   }
@@ -807,6 +825,41 @@ torch::Tensor conv2d_wgrad(torch::Tensor x, torch::Tensor dy, long R, long S,
         dw.data_ptr<float>(), g, NP, npslice);
   }
   return dw.view({(long)g.K, R, S, (long)g.C});
+}
+
+// wgrad accumulated straight into the fp32 OIHW flat-grad view (pre-zeroed
+// by the per-step zero_grad): no fresh allocation, no layout transform, no
+// autograd accumulate-add. Also serves Linear (x [M,1,1,Cin], dy
+// [M,1,1,Nout], R=S=1: OIHW == [Nout, Cin] row-major == torch weight grad).
+void conv2d_wgrad_into(torch::Tensor x, torch::Tensor dy, torch::Tensor dw,
+                       long R, long S, long stride, long pad) {
+  CHECK_IN(x); CHECK_IN(dy); CHECK_IN(dw);
+  TORCH_CHECK(dw.scalar_type() == torch::kFloat32);
+  auto g = make_geom(x, (int)dy.size(3), (int)R, (int)S, stride, pad);
+  TORCH_CHECK(g.Ho == (int)dy.size(1) && g.Wo == (int)dy.size(2),
+              "wgrad geometry mismatch");
+  TORCH_CHECK(dw.numel() == (long)g.K * g.C * R * S);
+  long NP = (long)g.N * g.Ho * g.Wo;
+  long RED = (long)R * S * g.C;
+  long target_blocks = 2048;
+  long tiles = (long)ceil_div(g.K, conv::WBM) * ceil_div(RED, conv::WBN);
+  long zsplit =
+      std::max(1L, std::min(512L, target_blocks / std::max(tiles, 1L)));
+  long npslice = (NP + zsplit - 1) / zsplit;
+  npslice = ((npslice + conv::WBK - 1) / conv::WBK) * conv::WBK;
+  zsplit = (NP + npslice - 1) / npslice;
+  bool fast = (g.K % 8 == 0) && (g.C % 8 == 0);
+  dim3 grid(ceil_div(g.K, conv::WBM), ceil_div(RED, conv::WBN),
+            (unsigned)zsplit);
+  if (fast) {
+    conv::conv_wgrad_fast_kernel<true, true><<<grid, 256, 0, cur_stream()>>>(
+        (const bf16*)x.data_ptr(), (const bf16*)dy.data_ptr(),
+        dw.data_ptr<float>(), g, NP, npslice);
+  } else {
+    conv::conv_wgrad_kernel<false, true><<<grid, 256, 0, cur_stream()>>>(
+        (const bf16*)x.data_ptr(), (const bf16*)dy.data_ptr(),
+        dw.data_ptr<float>(), g, NP, npslice);
+  }
 }
 
 }  // namespace eg

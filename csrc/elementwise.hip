@@ -185,6 +185,18 @@ torch::Tensor channel_sum(torch::Tensor x) {
   return out;
 }
 
+// accumulate the channel sum straight into a pre-zeroed flat-grad view
+// (bias grads written by the producing kernel; no aten accumulate-add)
+void channel_sum_into(torch::Tensor x, torch::Tensor out) {
+  CHECK_IN(x);
+  CHECK_IN(out);
+  int c = (int)x.size(-1);
+  long rows = x.numel() / c;
+  dim3 grid((c + 63) / 64, (unsigned)std::min<long>((rows + 3) / 4, 256L));
+  channel_sum_kernel<<<grid, 256, 0, cur_stream()>>>(
+      (const bf16*)x.data_ptr(), out.data_ptr<float>(), rows, c);
+}
+
 
 // ---- fused conv-weight layout transforms (one kernel per transform) ------
 // OIHW fp32 -> KRSC bf16 (fwd operand), KRSC bf16 -> CRSK bf16 (dgrad
@@ -281,6 +293,124 @@ torch::Tensor krsc_to_oihw(torch::Tensor dwk) {
   wt::krsc_to_oihw_kernel<<<wt_grid(total), 256, 0, cur_stream()>>>(
       dwk.data_ptr<float>(), out.data_ptr<float>(), K, C, R, S);
   return out;
+}
+
+// ---- persistent weight-shadow refresh ------------------------------------
+// ONE launch regenerates every conv/linear weight's bf16 KRSC (fwd operand)
+// and CRSK (dgrad operand) shadow from the updated fp32 flat params —
+// replacing the per-conv oihw_to_krsc + krsc_to_crsk launches on every
+// forward/backward. Runs eagerly right after the fused
+// SGD step (the last param mutation of a pass), so the hipGraph-captured
+// fwd+bwd reads shadows at fixed addresses with fresh values.
+//
+// Thread mapping: vec8 chunks over the PADDED shadow offset space
+// (sh_start[j] is 8-aligned); dst-indexed so shadow writes are coalesced
+// 16-B stores, the fp32 reads scatter (stride R*S or C*R*S) but stay
+// within one filter's footprint and hit L2.
+
+namespace wt {
+
+__global__ void refresh_shadows_kernel(
+    const float* __restrict__ param, const long* __restrict__ p_start,
+    const long* __restrict__ sh_start, const int* __restrict__ Ks,
+    const int* __restrict__ Cs, const int* __restrict__ Rs,
+    const int* __restrict__ Ss, int nseg, long total8,
+    bf16* __restrict__ krsc, bf16* __restrict__ crsk) {
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (long v = (long)blockIdx.x * blockDim.x + threadIdx.x; v < 2 * total8;
+       v += stride) {
+    const bool is_crsk = v >= total8;
+    const long e0 = (is_crsk ? v - total8 : v) * 8;  // padded shadow offset
+    int lo = 0, hi = nseg - 1;
+    while (lo < hi) {
+      int mid = (lo + hi + 1) >> 1;
+      if (sh_start[mid] <= e0) lo = mid; else hi = mid - 1;
+    }
+    const int j = lo;
+    const int K = Ks[j], C = Cs[j], R = Rs[j], S = Ss[j];
+    const long numel = (long)K * C * R * S;
+    const long loc = e0 - sh_start[j];
+    if (loc >= numel) continue;  // pad gap
+    const float* w = param + p_start[j];
+    __bf16* dst = reinterpret_cast<__bf16*>(
+        (is_crsk ? crsk : krsc) + e0);
+    const int RS = R * S, CRS = C * RS;
+    if (!is_crsk) {
+      // krsc idx loc+t = ((k*R+r)*S+s)*C + c  (c fastest)
+      const int c0 = (int)(loc % C);
+      long rem = loc / C;
+      const int s = (int)(rem % S);
+      rem /= S;
+      const int r = (int)(rem % R);
+      const int k = (int)(rem / R);
+      if (c0 + 8 <= C && loc + 8 <= numel) {
+        const float* src = w + (long)k * CRS + (long)c0 * RS + r * S + s;
+        bf16x8 o;
+#pragma unroll
+        for (int t = 0; t < 8; ++t) o[t] = (__bf16)src[(long)t * RS];
+        *reinterpret_cast<bf16x8*>(dst) = o;
+      } else {
+        for (int t = 0; t < 8 && loc + t < numel; ++t) {
+          long i = loc + t;
+          int c = (int)(i % C);
+          long rm = i / C;
+          int ss = (int)(rm % S);
+          rm /= S;
+          int rr = (int)(rm % R);
+          int kk = (int)(rm / R);
+          dst[t] = (__bf16)w[(long)kk * CRS + (long)c * RS + rr * S + ss];
+        }
+      }
+    } else {
+      // crsk idx loc+t = ((c*R+r)*S+s)*K + k  (k fastest)
+      const int k0 = (int)(loc % K);
+      long rem = loc / K;
+      const int s = (int)(rem % S);
+      rem /= S;
+      const int r = (int)(rem % R);
+      const int c = (int)(rem / R);
+      if (k0 + 8 <= K && loc + 8 <= numel) {
+        const float* src = w + (long)k0 * CRS + (long)c * RS + r * S + s;
+        bf16x8 o;
+#pragma unroll
+        for (int t = 0; t < 8; ++t) o[t] = (__bf16)src[(long)t * CRS];
+        *reinterpret_cast<bf16x8*>(dst) = o;
+      } else {
+        for (int t = 0; t < 8 && loc + t < numel; ++t) {
+          long i = loc + t;
+          int kk = (int)(i % K);
+          long rm = i / K;
+          int ss = (int)(rm % S);
+          rm /= S;
+          int rr = (int)(rm % R);
+          int cc = (int)(rm / R);
+          dst[t] = (__bf16)w[(long)kk * CRS + (long)cc * RS + rr * S + ss];
+        }
+      }
+    }
+  }
+}
+
+}  // namespace wt
+
+void refresh_conv_shadows(torch::Tensor param, torch::Tensor p_start,
+                          torch::Tensor sh_start, torch::Tensor Ks,
+                          torch::Tensor Cs, torch::Tensor Rs,
+                          torch::Tensor Ss, torch::Tensor krsc,
+                          torch::Tensor crsk) {
+  CHECK_IN(param);
+  CHECK_IN(p_start);
+  int nseg = (int)p_start.numel();
+  if (nseg == 0) return;
+  long total8 = krsc.numel() / 8;
+  TORCH_CHECK(krsc.numel() % 8 == 0 && crsk.numel() == krsc.numel());
+  long work = 2 * total8;
+  int grid = (int)std::min<long>((work + 255) / 256, 4096L);
+  wt::refresh_shadows_kernel<<<grid, 256, 0, cur_stream()>>>(
+      param.data_ptr<float>(), p_start.data_ptr<long>(),
+      sh_start.data_ptr<long>(), Ks.data_ptr<int>(), Cs.data_ptr<int>(),
+      Rs.data_ptr<int>(), Ss.data_ptr<int>(), nseg, total8,
+      (bf16*)krsc.data_ptr(), (bf16*)crsk.data_ptr());
 }
 
 }  // namespace eg

@@ -43,7 +43,9 @@ class Conv2d(nn.Module):
         # the paired BatchNorm downgrades stats_ready by the same condition
         return O.conv2d(x, self.weight, self.bias, self.stride, self.padding,
                         bn_stats=bn_stats and O.use_native(x) and self.training
-                        and self.out_ch % 64 == 0)
+                        and self.out_ch % 64 == 0,
+                        wk=getattr(self, "_shadow_wk", None),
+                        wt=getattr(self, "_shadow_wt", None))
 
     def extra_repr(self):
         return (f"{self.in_ch}, {self.out_ch}, k={self.kernel_size}, "
@@ -103,7 +105,9 @@ class Linear(nn.Module):
             nn.init.uniform_(self.bias, -bound, bound)
 
     def forward(self, x):
-        return O.linear(x, self.weight, self.bias)
+        return O.linear(x, self.weight, self.bias,
+                        wk=getattr(self, "_shadow_wk", None),
+                        wt=getattr(self, "_shadow_wt", None))
 
     def extra_repr(self):
         return f"{self.in_features}, {self.out_features}"

@@ -93,13 +93,28 @@ def _w_krsc(w: torch.Tensor) -> torch.Tensor:
 # --------------------------------------------------------------------------
 
 class _ConvNHWC(torch.autograd.Function):
+    """Direct-grad convention: when the weight's .grad
+    is a pre-allocated flat-space view (FlatParamSpace re-points it), the
+    wgrad kernel ACCUMULATES the OIHW gradient straight into that view and
+    backward returns None for the weight — no autograd accumulate-add, no
+    layout transform. Requires the per-step zero_grad of the flat buffer
+    and single use of each weight per step (true for the whole model zoo;
+    a reused weight would drop all but the kernels' own accumulation).
+    Standalone tensors (unit tests) take the allocate-and-return path.
+    """
+
     @staticmethod
-    def forward(ctx, x, w, bias, stride, padding, bn_stats):
+    def forward(ctx, x, w, bias, stride, padding, bn_stats, wk, wt):
         core = native()
-        wk = _w_krsc(w)
+        if wk is None:
+            wk = _w_krsc(w)
         b = bias.detach() if bias is not None else _empty_f32(x.device)
         y = core.conv2d_fwd(x, wk, b, stride, padding, bn_stats)
         ctx.save_for_backward(x, wk)
+        ctx.wt = wt                       # persistent CRSK shadow (or None)
+        ctx.wgrad = w.grad if w.requires_grad else None
+        ctx.bgrad = (bias.grad if (bias is not None and bias.requires_grad)
+                     else None)
         ctx.stride, ctx.padding = stride, padding
         ctx.has_bias = bias is not None
         ctx.hw = (x.shape[1], x.shape[2])
@@ -111,27 +126,38 @@ class _ConvNHWC(torch.autograd.Function):
         x, wk = ctx.saved_tensors
         dy = dy.contiguous()
         dx = dw = db = None
+        R, S = wk.shape[1], wk.shape[2]
         if ctx.needs_input_grad[0]:
             # dgrad weight layout: [C,R,S,K] from [K,R,S,C]
-            w_crsk = core.krsc_to_crsk(wk)
+            w_crsk = ctx.wt if ctx.wt is not None else core.krsc_to_crsk(wk)
             dx = core.conv2d_dgrad(dy, w_crsk, ctx.stride, ctx.padding,
                                    ctx.hw[0], ctx.hw[1])
         if ctx.needs_input_grad[1]:
-            dw_krsc = core.conv2d_wgrad(x, dy, wk.shape[1], wk.shape[2],
-                                        ctx.stride, ctx.padding)
-            dw = core.krsc_to_oihw(dw_krsc)  # KRSC fp32 -> OIHW fp32
+            if ctx.wgrad is not None:
+                core.conv2d_wgrad_into(x, dy, ctx.wgrad, R, S,
+                                       ctx.stride, ctx.padding)
+            else:
+                dw_krsc = core.conv2d_wgrad(x, dy, R, S, ctx.stride,
+                                            ctx.padding)
+                dw = core.krsc_to_oihw(dw_krsc)  # KRSC fp32 -> OIHW fp32
         if ctx.has_bias and ctx.needs_input_grad[2]:
-            db = core.channel_sum(dy)
-        return dx, dw, db, None, None, None
+            if ctx.bgrad is not None:
+                core.channel_sum_into(dy, ctx.bgrad)
+            else:
+                db = core.channel_sum(dy)
+        return dx, dw, db, None, None, None, None, None
 
 
-def conv2d(x, w, bias=None, stride=1, padding=0, bn_stats=False):
+def conv2d(x, w, bias=None, stride=1, padding=0, bn_stats=False,
+           wk=None, wt=None):
     """bn_stats: fuse the following training-mode BatchNorm's batch-stats
     accumulation into this conv's epilogue (pair with
-    batch_norm(..., stats_ready=True))."""
+    batch_norm(..., stats_ready=True)). wk/wt: persistent bf16 KRSC/CRSK
+    shadow views maintained by FlatParamSpace.refresh_shadows (skips the
+    per-use layout transforms)."""
     if use_native(x):
         return _ConvNHWC.apply(x, w, bias, int(stride), int(padding),
-                               bool(bn_stats))
+                               bool(bn_stats), wk, wt)
     return F.conv2d(x, w, bias, stride=stride, padding=padding)
 
 
@@ -148,6 +174,11 @@ class _BatchNormNHWC(torch.autograd.Function):
             x, gamma.detach(), beta.detach(), running_mean, running_var,
             momentum, eps, training, fuse_relu, stats_ready)
         ctx.save_for_backward(x, gamma, save_mean, save_invstd, y)
+        # direct-grad views (see _ConvNHWC): the bwd reduction buffers ARE
+        # dgamma/dbeta, so pointing them at the pre-zeroed flat-grad views
+        # writes the gradients in place with zero extra kernels
+        ctx.ggrad = gamma.grad if gamma.requires_grad else None
+        ctx.bgrad = beta.grad if beta.requires_grad else None
         ctx.fuse_relu = fuse_relu
         ctx.training = training
         return y
@@ -156,6 +187,16 @@ class _BatchNormNHWC(torch.autograd.Function):
     def backward(ctx, dy):
         core = native()
         x, gamma, save_mean, save_invstd, y = ctx.saved_tensors
+        direct = (ctx.ggrad is not None and ctx.bgrad is not None
+                  and ctx.needs_input_grad[1] and ctx.needs_input_grad[2])
+        if direct:
+            dx, _, _ = core.bn_bwd(dy.contiguous(), x, save_mean,
+                                   save_invstd, gamma.detach(), y,
+                                   ctx.fuse_relu, ctx.training,
+                                   dgamma_out=ctx.ggrad,
+                                   dbeta_out=ctx.bgrad)
+            return (dx, None, None, None, None, None, None, None, None,
+                    None)
         dx, dgamma, dbeta = core.bn_bwd(dy.contiguous(), x, save_mean,
                                         save_invstd, gamma.detach(), y,
                                         ctx.fuse_relu, ctx.training)
@@ -303,15 +344,23 @@ def avg_pool(x, k):
 
 class _Linear(torch.autograd.Function):
     """GEMM is NT form: gemm_bias(A[M,K], B[N,K]) = A @ B.T — torch Linear
-    weight [N,K] feeds the forward without any transpose."""
+    weight [N,K] feeds the forward without any transpose. With shadow views
+    (wk = bf16 [N,K], wt = bf16 [K,N]) the per-use cast and the backward
+    transpose copies disappear; with a flat-space .grad view the weight
+    gradient is accumulated in place by the TN wgrad kernel (dw[N,K] =
+    dy^T @ x == a 1x1 conv wgrad over [M,1,1,*] views)."""
 
     @staticmethod
-    def forward(ctx, x, w, bias):
+    def forward(ctx, x, w, bias, wk, wt):
         core = native()
-        wb = w.detach().to(torch.bfloat16)                   # [N, K]
+        wb = wk if wk is not None else w.detach().to(torch.bfloat16)
         b = bias.detach() if bias is not None else _empty_f32(x.device)
         y = core.gemm_bias(x, wb, b, True)
         ctx.save_for_backward(x, wb)
+        ctx.wt = wt
+        ctx.wgrad = w.grad if w.requires_grad else None
+        ctx.bgrad = (bias.grad if (bias is not None and bias.requires_grad)
+                     else None)
         ctx.has_bias = bias is not None
         return y
 
@@ -323,20 +372,30 @@ class _Linear(torch.autograd.Function):
         dx = dw = db = None
         e = _empty_f32(x.device)
         if ctx.needs_input_grad[0]:
-            wt = wb.t().contiguous()                         # [K, N]
+            wt = ctx.wt if ctx.wt is not None else wb.t().contiguous()
             dx = core.gemm_bias(dy, wt, e, True)
         if ctx.needs_input_grad[1]:
-            # dw[N,K] = dy^T[N,M] @ (x^T[K,M])^T, fp32 out
-            dw = core.gemm_bias(dy.t().contiguous(),
-                                x.t().contiguous(), e, False)
+            if ctx.wgrad is not None:
+                m, k = x.shape
+                n = dy.shape[1]
+                core.conv2d_wgrad_into(x.view(m, 1, 1, k),
+                                       dy.view(m, 1, 1, n),
+                                       ctx.wgrad, 1, 1, 1, 0)
+            else:
+                # dw[N,K] = dy^T[N,M] @ (x^T[K,M])^T, fp32 out
+                dw = core.gemm_bias(dy.t().contiguous(),
+                                    x.t().contiguous(), e, False)
         if ctx.has_bias and ctx.needs_input_grad[2]:
-            db = core.channel_sum(dy)
-        return dx, dw, db
+            if ctx.bgrad is not None:
+                core.channel_sum_into(dy, ctx.bgrad)
+            else:
+                db = core.channel_sum(dy)
+        return dx, dw, db, None, None
 
 
-def linear(x, w, bias=None):
+def linear(x, w, bias=None, wk=None, wt=None):
     if use_native(x):
-        return _Linear.apply(x, w, bias)
+        return _Linear.apply(x, w, bias, wk, wt)
     return F.linear(x, w, bias)
 
 

@@ -168,6 +168,9 @@ class CommEngine:
         o = self.cfg.optim
         self._last_norms_sq = self.K.sgd_step(self.space, o.lr, o.momentum,
                                               o.weight_decay)
+        # the step is the last param mutation of a pass: regenerate the
+        # bf16 weight shadows the next forward/backward will read
+        self.space.refresh_shadows()
 
     def finalize(self) -> dict:
         """Closing consensus: params <- allreduce/world (event.cpp:517-525).
@@ -178,6 +181,7 @@ class CommEngine:
         if self.world > 1:
             dist.all_reduce(self.space.param)
             self.space.param.div_(self.world)
+            self.space.refresh_shadows()  # post-consensus eval reads them
             # NCCL needs a device tensor (gloo accepts either)
             ev = torch.tensor([self.num_events], dtype=torch.int64,
                               device=self.device)

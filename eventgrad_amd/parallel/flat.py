@@ -69,6 +69,85 @@ class FlatParamSpace:
             p.data = view
             p.grad = self.grad[s:s + n].view(shape)
 
+        self._build_shadows(model, params)
+
+    # -- persistent bf16 weight shadows (GPU native path) ------------------
+    # Every conv/linear weight keeps two bf16 shadow copies in kernel-ready
+    # layouts: KRSC (conv fwd operand / linear [N,K]) and CRSK (conv dgrad
+    # operand / linear [K,N] transpose). ONE kernel launch regenerates all
+    # of them from the fp32 flat params after each mutation (fused SGD
+    # step, consensus allreduce, checkpoint load), replacing the per-conv
+    # per-step oihw_to_krsc / krsc_to_crsk transform launches inside the
+    # captured fwd+bwd graph.
+    def _build_shadows(self, model, params) -> None:
+        self._shadow_meta = None
+        if self.device.type != "cuda":
+            return
+        from ..models.layers import Conv2d as EgConv2d
+        from ..models.layers import Linear as EgLinear
+
+        by_param = {id(p): i for i, p in enumerate(params)}
+        entries = []  # (seg_idx, K, C, R, S, module)
+        for mod in model.modules():
+            if isinstance(mod, EgConv2d):
+                i = by_param.get(id(mod.weight))
+                if i is None:
+                    continue
+                K, C, R, S = self.shapes[i]
+                entries.append((i, K, C, R, S, mod))
+            elif isinstance(mod, EgLinear):
+                i = by_param.get(id(mod.weight))
+                if i is None:
+                    continue
+                N, Kin = self.shapes[i]
+                entries.append((i, N, Kin, 1, 1, mod))
+        if not entries:
+            return
+        off = 0
+        sh_starts = []
+        for (i, K, C, R, S, mod) in entries:
+            sh_starts.append(off)
+            off += (K * C * R * S + ALIGN - 1) // ALIGN * ALIGN
+        dev = self.device
+        bf16 = dict(dtype=torch.bfloat16, device=dev)
+        self.shadow_krsc = torch.zeros(off, **bf16)
+        self.shadow_crsk = torch.zeros(off, **bf16)
+        i64 = dict(dtype=torch.int64, device=dev)
+        i32 = dict(dtype=torch.int32, device=dev)
+        self._sh_pstart = torch.tensor([self.starts[e[0]] for e in entries],
+                                       **i64)
+        self._sh_start = torch.tensor(sh_starts, **i64)
+        self._sh_K = torch.tensor([e[1] for e in entries], **i32)
+        self._sh_C = torch.tensor([e[2] for e in entries], **i32)
+        self._sh_R = torch.tensor([e[3] for e in entries], **i32)
+        self._sh_S = torch.tensor([e[4] for e in entries], **i32)
+        for (i, K, C, R, S, mod), so in zip(entries, sh_starts):
+            n = K * C * R * S
+            wk = self.shadow_krsc[so:so + n]
+            wt = self.shadow_crsk[so:so + n]
+            if isinstance(mod, EgConv2d):
+                mod._shadow_wk = wk.view(K, R, S, C)
+                mod._shadow_wt = wt.view(C, R, S, K)
+            else:  # Linear: [N, Kin] and its transpose
+                mod._shadow_wk = wk.view(K, C)
+                mod._shadow_wt = wt.view(C, K)
+        self._shadow_meta = True
+        self.refresh_shadows()
+
+    def refresh_shadows(self) -> None:
+        """Regenerate every weight shadow from the fp32 flat params (one
+        kernel). Call after any param mutation outside fwd/bwd."""
+        if self._shadow_meta is None:
+            return
+        from ..ops import functional as O
+        if O.get_compute_dtype() != "bf16":
+            return
+        from ..ops.backend import native
+        native().refresh_conv_shadows(
+            self.param, self._sh_pstart, self._sh_start, self._sh_K,
+            self._sh_C, self._sh_R, self._sh_S, self.shadow_krsc,
+            self.shadow_crsk)
+
     # -- views ------------------------------------------------------------
     def seg(self, buf: torch.Tensor, i: int) -> torch.Tensor:
         s, n = self.starts[i], self.numels[i]
@@ -91,3 +170,4 @@ class FlatParamSpace:
     def load_flat(self, flat: torch.Tensor) -> None:
         with torch.no_grad():
             self.param.copy_(flat)
+        self.refresh_shadows()

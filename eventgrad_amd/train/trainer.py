@@ -70,6 +70,7 @@ class Trainer:
                                  self.device, model=self.model)
             self.pass_num = st["pass_num"]
             self.start_epoch = st["epoch"] + 1
+            self.space.refresh_shadows()
 
     # ------------------------------------------------------------------
     def train(self) -> RunMetrics:
