@@ -50,7 +50,8 @@ std::vector<torch::Tensor> bn_fwd(torch::Tensor, torch::Tensor, torch::Tensor,
                                   double, bool, bool, bool);
 std::vector<torch::Tensor> bn_bwd(torch::Tensor, torch::Tensor, torch::Tensor,
                                   torch::Tensor, torch::Tensor, torch::Tensor,
-                                  bool, bool, torch::Tensor, torch::Tensor);
+                                  bool, bool, c10::optional<torch::Tensor>,
+                                  c10::optional<torch::Tensor>);
 // pool.hip
 std::vector<torch::Tensor> maxpool2x2_fwd(torch::Tensor);
 torch::Tensor maxpool2x2_bwd(torch::Tensor, torch::Tensor, long, long);
@@ -102,8 +103,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("have_stats") = false);
   m.def("bn_bwd", &eg::bn_bwd, py::arg("dy"), py::arg("x"), py::arg("mean"),
         py::arg("invstd"), py::arg("gamma"), py::arg("y"), py::arg("relu"),
-        py::arg("training"), py::arg("dgamma_out") = torch::Tensor(),
-        py::arg("dbeta_out") = torch::Tensor());
+        py::arg("training"),
+        py::arg("dgamma_out") = c10::nullopt,
+        py::arg("dbeta_out") = c10::nullopt);
   m.def("maxpool2x2_fwd", &eg::maxpool2x2_fwd);
   m.def("maxpool2x2_bwd", &eg::maxpool2x2_bwd);
   m.def("avgpool_fwd", &eg::avgpool_fwd);

@@ -395,8 +395,8 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor x,
                                   torch::Tensor mean, torch::Tensor invstd,
                                   torch::Tensor gamma, torch::Tensor y,
                                   bool relu, bool training,
-                                  torch::Tensor dgamma_out,
-                                  torch::Tensor dbeta_out) {
+                                  c10::optional<torch::Tensor> dgamma_out,
+                                  c10::optional<torch::Tensor> dbeta_out) {
   CHECK_IN(dy); CHECK_IN(x);
   int c = (int)x.size(-1);
   long rows = x.numel() / c;
@@ -407,11 +407,11 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor x,
   // accumulates the gradients in place — no allocation, no aten
   // accumulate-add afterwards.
   torch::Tensor sum_dy, sum_dyx;
-  const bool direct = dgamma_out.defined() && dgamma_out.numel() == c;
+  const bool direct = dgamma_out.has_value() && dgamma_out->numel() == c;
   if (direct) {
-    CHECK_IN(dgamma_out); CHECK_IN(dbeta_out);
-    sum_dyx = dgamma_out;
-    sum_dy = dbeta_out;
+    CHECK_IN(*dgamma_out); CHECK_IN(*dbeta_out);
+    sum_dyx = *dgamma_out;
+    sum_dy = *dbeta_out;
   } else {
     auto sums = torch::zeros({2 * c}, f32);
     sum_dy = sums.narrow(0, 0, c);
