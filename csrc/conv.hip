@@ -349,13 +349,7 @@ __device__ __forceinline__ bf16x8 tr16_frag(const __bf16* base, int np0,
 __device__ __forceinline__ int np_img(int np) { return np * 16 + (np >> 3) * 8; }
 constexpr int WIMG = WBK * 16 + (WBK / 8) * 8 + 16;
 
-// OIHW: write dw straight into the fp32 OIHW flat-grad view (accumulate
-// semantics into the pre-zeroed grad buffer) instead of a fresh [K, RSC]
-// KRSC tensor — removes the krsc_to_oihw transform AND the autograd
-// accumulate-add from the per-step path. The rsc
-// column decomposes once per fragment column; the OIHW offset is then
-// kk * CRS + (c*R + r)*S + s.
-template <bool ATOMIC, bool OIHW = false>
+template <bool ATOMIC>
 __global__ __launch_bounds__(256) void conv_wgrad_fast_kernel(
     const bf16* __restrict__ x, const bf16* __restrict__ dy,
     float* __restrict__ dw, Geom g, long NP, long npslice) {
@@ -488,27 +482,21 @@ __global__ __launch_bounds__(256) void conv_wgrad_fast_kernel(
     for (int fj = 0; fj < 2; ++fj) {
       long nn = n0 + wc * 32 + fj * 16 + cn;
       if (nn >= RED) continue;
-      long col = nn;
-      if (OIHW) {
-        const int c = (int)(nn % g.C);
-        const int rs = (int)(nn / g.C);
-        col = ((long)c * g.R + rs / g.S) * g.S + rs % g.S;
-      }
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         long kk = k0c + wr * 32 + fi * 16 + cm + r;
         if (kk >= g.K) continue;
         if (ATOMIC) {
-          atomicAdd(&dw[kk * RED + col], acc[fi][fj][r]);
+          atomicAdd(&dw[kk * RED + nn], acc[fi][fj][r]);
         } else {
-          dw[kk * RED + col] = acc[fi][fj][r];
+          dw[kk * RED + nn] = acc[fi][fj][r];
         }
       }
     }
   }
 }
 
-template <bool FAST, bool OIHW = false>  // generic fallback (any C/K)
+template <bool FAST>  // generic fallback (any C/K); WBK_GEN=32 transposed
 __global__ __launch_bounds__(256) void conv_wgrad_kernel(
     const bf16* __restrict__ x, const bf16* __restrict__ dy,
     float* __restrict__ dw, Geom g, long NP, long npslice) {
@@ -644,17 +632,11 @@ __global__ __launch_bounds__(256) void conv_wgrad_kernel(
     for (int fj = 0; fj < 2; ++fj) {
       long nn = n0 + wc * 32 + fj * 16 + cn;
       if (nn >= RED) continue;
-      long col = nn;
-      if (OIHW) {
-        const int c = (int)(nn % g.C);
-        const int rs = (int)(nn / g.C);
-        col = ((long)c * g.R + rs / g.S) * g.S + rs % g.S;
-      }
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         long kk = k0c + wr * 32 + fi * 16 + cm + r;
         if (kk >= g.K) continue;
-        atomicAdd(&dw[kk * RED + col], acc[fi][fj][r]);
+        atomicAdd(&dw[kk * RED + nn], acc[fi][fj][r]);
       }
     }
   }
@@ -827,10 +809,52 @@ torch::Tensor conv2d_wgrad(torch::Tensor x, torch::Tensor dy, long R, long S,
   return dw.view({(long)g.K, R, S, (long)g.C});
 }
 
-// wgrad accumulated straight into the fp32 OIHW flat-grad view (pre-zeroed
-// by the per-step zero_grad): no fresh allocation, no layout transform, no
-// autograd accumulate-add. Also serves Linear (x [M,1,1,Cin], dy
-// [M,1,1,Nout], R=S=1: OIHW == [Nout, Cin] row-major == torch weight grad).
+// persistent KRSC wgrad scratch, shared by every conv: allocated zeroed
+// once and RETURNED to zero by the consuming transform kernel (same
+// recycling trick as the BN stats workspace), so the per-conv zero-fill
+// launches disappear. Grows monotonically; stable after the first pass,
+// so hipGraph capture sees a fixed address.
+static torch::Tensor& wgrad_scratch(long n, const torch::TensorOptions& f32) {
+  static auto* t = new torch::Tensor();
+  if (!t->defined() || t->numel() < n) *t = torch::zeros({n}, f32);
+  return *t;
+}
+
+namespace conv {
+
+// grad_oihw[i] += scratch_krsc[perm(i)]; scratch reset to 0 after the read.
+// dst-indexed (coalesced grad writes + scattered scratch reads, same shape
+// as the old krsc_to_oihw transform that measured ~6 us per conv).
+__global__ void krsc_accum_oihw_reset_kernel(float* __restrict__ scratch,
+                                             float* __restrict__ grad, int K,
+                                             int C, int R, int S) {
+  long total = (long)K * C * R * S;
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += stride) {
+    // OIHW i -> (k, c, r, s) -> KRSC index
+    int s = (int)(i % S);
+    long rem = i / S;
+    int r = (int)(rem % R);
+    rem /= R;
+    int c = (int)(rem % C);
+    int k = (int)(rem / C);
+    long j = ((long)(k * R + r) * S + s) * C + c;
+    grad[i] += scratch[j];
+    scratch[j] = 0.f;
+  }
+}
+
+}  // namespace conv
+
+// wgrad accumulated into the fp32 OIHW flat-grad view (pre-zeroed by the
+// per-step zero_grad): the MFMA kernel keeps its fast coalesced KRSC
+// epilogue (an OIHW epilogue measured 4x slower: the rsc->hw permutation
+// scatters the fp32 atomics across cachelines), writing into the shared
+// self-rezeroing scratch; one cheap transform then folds scratch into the
+// grad view. No fresh allocation, no zero fill, no autograd
+// accumulate-add. Also serves Linear (x [M,1,1,Cin], dy [M,1,1,Nout],
+// R=S=1: OIHW == [Nout, Cin] row-major == the torch weight-grad layout).
 void conv2d_wgrad_into(torch::Tensor x, torch::Tensor dy, torch::Tensor dw,
                        long R, long S, long stride, long pad) {
   CHECK_IN(x); CHECK_IN(dy); CHECK_IN(dw);
@@ -838,9 +862,10 @@ void conv2d_wgrad_into(torch::Tensor x, torch::Tensor dy, torch::Tensor dw,
   auto g = make_geom(x, (int)dy.size(3), (int)R, (int)S, stride, pad);
   TORCH_CHECK(g.Ho == (int)dy.size(1) && g.Wo == (int)dy.size(2),
               "wgrad geometry mismatch");
-  TORCH_CHECK(dw.numel() == (long)g.K * g.C * R * S);
-  long NP = (long)g.N * g.Ho * g.Wo;
   long RED = (long)R * S * g.C;
+  long total = (long)g.K * RED;
+  TORCH_CHECK(dw.numel() == total);
+  long NP = (long)g.N * g.Ho * g.Wo;
   long target_blocks = 2048;
   long tiles = (long)ceil_div(g.K, conv::WBM) * ceil_div(RED, conv::WBN);
   long zsplit =
@@ -851,15 +876,21 @@ void conv2d_wgrad_into(torch::Tensor x, torch::Tensor dy, torch::Tensor dw,
   bool fast = (g.K % 8 == 0) && (g.C % 8 == 0);
   dim3 grid(ceil_div(g.K, conv::WBM), ceil_div(RED, conv::WBN),
             (unsigned)zsplit);
+  auto& scratch = wgrad_scratch(total, x.options().dtype(torch::kFloat32));
+  float* sp = scratch.data_ptr<float>();
+  // scratch is zero here (invariant); atomics accumulate, transform resets
   if (fast) {
-    conv::conv_wgrad_fast_kernel<true, true><<<grid, 256, 0, cur_stream()>>>(
-        (const bf16*)x.data_ptr(), (const bf16*)dy.data_ptr(),
-        dw.data_ptr<float>(), g, NP, npslice);
+    conv::conv_wgrad_fast_kernel<true><<<grid, 256, 0, cur_stream()>>>(
+        (const bf16*)x.data_ptr(), (const bf16*)dy.data_ptr(), sp, g, NP,
+        npslice);
   } else {
-    conv::conv_wgrad_kernel<false, true><<<grid, 256, 0, cur_stream()>>>(
-        (const bf16*)x.data_ptr(), (const bf16*)dy.data_ptr(),
-        dw.data_ptr<float>(), g, NP, npslice);
+    conv::conv_wgrad_kernel<false><<<grid, 256, 0, cur_stream()>>>(
+        (const bf16*)x.data_ptr(), (const bf16*)dy.data_ptr(), sp, g, NP,
+        npslice);
   }
+  int tgrid = (int)std::min<long>((total + 255) / 256, 2048L);
+  conv::krsc_accum_oihw_reset_kernel<<<tgrid, 256, 0, cur_stream()>>>(
+      sp, dw.data_ptr<float>(), g.K, g.C, (int)R, (int)S);
 }
 
 }  // namespace eg

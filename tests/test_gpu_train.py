@@ -296,35 +296,51 @@ def test_shadow_views_match_transforms():
     check()
 
 
-def test_direct_grads_match_fallback():
+@pytest.mark.parametrize("model_name,mode", [("lenet5", "train"),
+                                             ("resnet20", "eval")])
+def test_direct_grads_match_fallback(model_name, mode):
     """Flat-space direct-grad writes (wgrad_into/bn outs/channel_sum_into)
-    must equal the standalone allocate-and-return autograd path."""
+    must equal the standalone allocate-and-return autograd path.
+
+    Forward must be deterministic for a strict comparison, so BN models run
+    in eval mode: train-mode BN batch stats are fp32 atomic reductions whose
+    ordering differs run-to-run, flipping last bf16 bits that amplify over
+    20 layers (measured ~20% per-tensor grad spread between two *identical*
+    train-mode runs). Eval-BN still exercises the direct dgamma/dbeta and
+    wgrad paths; lenet5 covers train mode (conv bias + linear, no BN).
+    """
     from eventgrad_amd.models import build_model
     from eventgrad_amd.ops import functional as O
     from eventgrad_amd.parallel.flat import FlatParamSpace
 
     dev = torch.device("cuda")
     torch.manual_seed(5)
-    m_ref = build_model("resnet20").to(dev)     # standalone: fallback path
+    m_ref = build_model(model_name).to(dev)     # standalone: fallback path
     torch.manual_seed(5)
-    m_flat = build_model("resnet20").to(dev)
+    m_flat = build_model(model_name).to(dev)
     space = FlatParamSpace(m_flat, dev)         # direct path
 
     torch.manual_seed(6)
     x = torch.randn(16, 3, 32, 32, device=dev)
     y = torch.randint(0, 10, (16,), device=dev)
 
-    m_ref.train(); m_flat.train()
-    loss_r = O.nll_of_logits(m_ref(x), y)
+    for m in (m_ref, m_flat):
+        m.train() if mode == "train" else m.eval()
+    torch.manual_seed(77)  # identical dropout seed draws in both runs
+    out_r = m_ref(x)
+    loss_r = O.nll_of_logits(out_r, y)
     loss_r.backward()
     space.zero_grad()
-    loss_f = O.nll_of_logits(m_flat(x), y)
+    torch.manual_seed(77)
+    out_f = m_flat(x)
+    loss_f = O.nll_of_logits(out_f, y)
     loss_f.backward()
     torch.cuda.synchronize()
-    # fp32 atomic orderings (BN stats epilogue) may flip last bits
-    assert torch.allclose(loss_r.detach(), loss_f.detach(), rtol=1e-3)
+    assert torch.equal(loss_r.detach(), loss_f.detach())
     for (name, pr), pf in zip(m_ref.named_parameters(), m_flat.parameters()):
         gr, gf = pr.grad, pf.grad
         denom = gr.norm().item() + 1e-12
+        # both paths run the same fp32 atomic reductions; only summation
+        # order differs
         err = (gr - gf).norm().item() / denom
         assert err < 1e-4, (name, err)
