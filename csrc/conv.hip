@@ -876,21 +876,29 @@ void conv2d_wgrad_into(torch::Tensor x, torch::Tensor dy, torch::Tensor dw,
   bool fast = (g.K % 8 == 0) && (g.C % 8 == 0);
   dim3 grid(ceil_div(g.K, conv::WBM), ceil_div(RED, conv::WBN),
             (unsigned)zsplit);
-  auto& scratch = wgrad_scratch(total, x.options().dtype(torch::kFloat32));
-  float* sp = scratch.data_ptr<float>();
-  // scratch is zero here (invariant); atomics accumulate, transform resets
+  // 1x1 (and Linear): KRSC == OIHW, so the epilogue atomics accumulate
+  // straight into the pre-zeroed grad view — no scratch, no fold kernel
+  float* dst = dw.data_ptr<float>();
+  const bool direct = (R == 1 && S == 1);
+  if (!direct) {
+    auto& scratch = wgrad_scratch(total, x.options().dtype(torch::kFloat32));
+    dst = scratch.data_ptr<float>();
+  }
+  // scratch is zero here (invariant); atomics accumulate, the fold resets
   if (fast) {
     conv::conv_wgrad_fast_kernel<true><<<grid, 256, 0, cur_stream()>>>(
-        (const bf16*)x.data_ptr(), (const bf16*)dy.data_ptr(), sp, g, NP,
+        (const bf16*)x.data_ptr(), (const bf16*)dy.data_ptr(), dst, g, NP,
         npslice);
   } else {
     conv::conv_wgrad_kernel<false><<<grid, 256, 0, cur_stream()>>>(
-        (const bf16*)x.data_ptr(), (const bf16*)dy.data_ptr(), sp, g, NP,
+        (const bf16*)x.data_ptr(), (const bf16*)dy.data_ptr(), dst, g, NP,
         npslice);
   }
-  int tgrid = (int)std::min<long>((total + 255) / 256, 2048L);
-  conv::krsc_accum_oihw_reset_kernel<<<tgrid, 256, 0, cur_stream()>>>(
-      sp, dw.data_ptr<float>(), g.K, g.C, (int)R, (int)S);
+  if (!direct) {
+    int tgrid = (int)std::min<long>((total + 255) / 256, 2048L);
+    conv::krsc_accum_oihw_reset_kernel<<<tgrid, 256, 0, cur_stream()>>>(
+        dst, dw.data_ptr<float>(), g.K, g.C, (int)R, (int)S);
+  }
 }
 
 }  // namespace eg
