@@ -45,6 +45,8 @@ def build_cfg(experiment: str, mode: str, args):
         cfg.epochs = args.epochs or 10
     cfg.data.synthetic_test_samples = 1024
     cfg.data.synthetic_noise = args.noise
+    cfg.data.label_noise = args.label_noise
+    cfg.seed = args.seed
     cfg.mode = mode
     cfg.trigger.adaptive = not args.constant
     cfg.trigger.horizon = args.horizon
@@ -89,6 +91,10 @@ def main():
     ap.add_argument("--epochs", type=int, default=None)
     ap.add_argument("--train-samples", type=int, default=None)
     ap.add_argument("--noise", type=float, default=0.8)
+    ap.add_argument("--label-noise", type=float, default=0.0,
+                    help="fraction of train labels flipped (test clean); "
+                         "holds accuracy in a non-saturating band")
+    ap.add_argument("--seed", type=int, default=0)
     ap.add_argument("--horizon", type=float, default=1.01)
     ap.add_argument("--constant", action="store_true",
                     help="use static threshold instead of adaptive")

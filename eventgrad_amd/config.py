@@ -71,6 +71,10 @@ class DataConfig:
     synthetic_test_samples: int = 10000
     num_classes: int = 10
     synthetic_noise: float = 0.5  # class-prototype noise level
+    # fraction of TRAIN labels flipped to a wrong class (test stays clean);
+    # keeps accuracy in a non-saturating band for discriminative
+    # matched-accuracy comparisons
+    label_noise: float = 0.0
 
 
 @dataclass

@@ -29,9 +29,19 @@ class SyntheticImages:
 
     def __init__(self, shape: Tuple[int, int, int], n: int,
                  num_classes: int = 10, noise: float = 0.5, seed: int = 1234,
-                 prototype_seed: int = 977):
+                 prototype_seed: int = 977, label_noise: float = 0.0):
         self.shape, self.n, self.num_classes = shape, n, num_classes
         self.noise = noise
+        # label noise: a deterministic fraction of samples gets a wrong
+        # label (uniform over the other classes) — used to hold test
+        # accuracy in a non-saturating band so matched-accuracy claims are
+        # discriminative
+        self.labels = np.arange(n, dtype=np.int64) % num_classes
+        if label_noise > 0:
+            lg = np.random.default_rng(seed * 31 + 5)
+            flip = lg.random(n) < label_noise
+            self.labels[flip] = (self.labels[flip] + lg.integers(
+                1, num_classes, flip.sum())) % num_classes
         # prototypes are shared between train/test splits (same
         # prototype_seed); only the per-index noise differs via `seed`.
         g = torch.Generator().manual_seed(prototype_seed)
@@ -63,7 +73,7 @@ class SyntheticImages:
     def __getitem__(self, i: int):
         i = int(i)
         return self._block(i // self.BLOCK)[i % self.BLOCK], \
-            i % self.num_classes
+            int(self.labels[i])
 
     def batch(self, idx: list) -> Tuple[torch.Tensor, torch.Tensor]:
         idx = np.asarray(idx, dtype=np.int64)
@@ -72,7 +82,7 @@ class SyntheticImages:
         for b in np.unique(blocks):
             sel = np.nonzero(blocks == b)[0]
             xs[sel] = self._block(int(b))[idx[sel] % self.BLOCK]
-        ys = torch.from_numpy(idx % self.num_classes)
+        ys = torch.from_numpy(self.labels[idx])
         return xs, ys
 
 
@@ -175,4 +185,5 @@ def build_dataset(cfg: DataConfig, train: bool):
         shape = _SHAPES["mnist"]
     n = cfg.synthetic_train_samples if train else cfg.synthetic_test_samples
     return SyntheticImages(shape, n, cfg.num_classes, cfg.synthetic_noise,
-                           seed=1234 if train else 4321)
+                           seed=1234 if train else 4321,
+                           label_noise=cfg.label_noise if train else 0.0)
