@@ -18,8 +18,10 @@ from __future__ import annotations
 import argparse
 import json
 import os
+import sys
 
 HERE = os.path.dirname(os.path.abspath(__file__))
+sys.path.insert(0, os.path.dirname(HERE))
 
 
 def run(dtype: str, args) -> dict:
