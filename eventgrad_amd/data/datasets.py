@@ -171,14 +171,83 @@ class Cifar10Dataset:
         return x, y
 
 
+# dcifar10/common/custom.hpp:15-19 labels map (CIFAR-10-images repo layout)
+CIFAR10_CLASSES = ("airplane", "automobile", "bird", "cat", "deer",
+                   "dog", "frog", "horse", "ship", "truck")
+
+
+class JpegDirCifar10:
+    """The reference's CIFAR-10 JPEG-directory contract
+    (dcifar10/common/custom.hpp:26-122): ``<root>/{train,test}/<class>/
+    0000.jpg ...`` with 5000 train / 1000 test images per class, decoded
+    to float CHW in R,G,B channel order. The reference resizes to 32x32
+    (cv::resize, custom.hpp:41) and keeps raw 0..255 float values — no
+    normalization anywhere in its pipeline — so this class matches that
+    exactly. Decoded images are cached as uint8 (~150 MB for the full
+    set). Shuffling is the sampler's job (the reference's
+    std::random_shuffle of the path list, custom.hpp:118-119, is replaced
+    by DistributedRandomSampler)."""
+
+    def __init__(self, root: str, train: bool = True, image_size: int = 32):
+        split = "train" if train else "test"
+        per_class = 5000 if train else 1000
+        self.image_size = image_size
+        self.files = []
+        labels = []
+        for j, cls in enumerate(CIFAR10_CLASSES):
+            d = os.path.join(root, split, cls)
+            if not os.path.isdir(d):
+                raise FileNotFoundError(f"missing class dir: {d}")
+            # the reference hardcodes 0000.jpg..NNNN.jpg; accept whatever
+            # count exists but keep the zero-padded name order
+            have = sorted(f for f in os.listdir(d) if f.endswith(".jpg"))
+            for f in have[:per_class]:
+                self.files.append(os.path.join(d, f))
+                labels.append(j)
+        self.labels = np.asarray(labels, dtype=np.int64)
+        self._cache: dict = {}
+
+    def __len__(self):
+        return len(self.files)
+
+    def _decode(self, i: int) -> np.ndarray:
+        arr = self._cache.get(i)
+        if arr is None:
+            from PIL import Image
+            im = Image.open(self.files[i]).convert("RGB")
+            if im.size != (self.image_size, self.image_size):
+                im = im.resize((self.image_size, self.image_size),
+                               Image.BILINEAR)  # cv::resize default
+            arr = np.asarray(im, dtype=np.uint8).transpose(2, 0, 1)  # CHW RGB
+            self._cache[i] = arr
+        return arr
+
+    def __getitem__(self, i: int):
+        # raw 0..255 float, like the reference (no normalization)
+        return torch.from_numpy(self._decode(i).astype(np.float32)), \
+            int(self.labels[i])
+
+    def batch(self, idx: list):
+        idx = np.asarray(idx, dtype=np.int64)
+        x = torch.from_numpy(np.stack([self._decode(int(i)) for i in idx])
+                             .astype(np.float32))
+        y = torch.from_numpy(self.labels[idx])
+        return x, y
+
+
 _SHAPES = {"mnist": (1, 28, 28), "cifar10": (3, 32, 32)}
 
 
 def build_dataset(cfg: DataConfig, train: bool):
     name = cfg.dataset
     if name in ("mnist", "cifar10") and cfg.data_path:
-        cls = MnistDataset if name == "mnist" else Cifar10Dataset
-        return cls(cfg.data_path, train)
+        if name == "mnist":
+            return MnistDataset(cfg.data_path, train)
+        # auto-detect the reference's JPEG-dir layout vs binary/pickle
+        if os.path.isdir(os.path.join(cfg.data_path, "train",
+                                      CIFAR10_CLASSES[0])):
+            return JpegDirCifar10(cfg.data_path, train)
+        return Cifar10Dataset(cfg.data_path, train)
     # synthetic fallback (no-network environment)
     shape = _SHAPES.get(name, _SHAPES["cifar10"])
     if name == "synthetic-mnist":

@@ -139,3 +139,47 @@ def test_cli_compute_dtype_flag():
     from eventgrad_amd.train.cli import build_parser, config_from_args
     args = build_parser().parse_args(["--compute-dtype", "fp32"])
     assert config_from_args(args).compute_dtype == "fp32"
+
+
+def test_jpeg_dir_cifar10(tmp_path):
+    """Reference JPEG-dir contract (dcifar10/common/custom.hpp:26-122):
+    train/<class>/NNNN.jpg tree, RGB CHW float 0..255, labels by dir."""
+    import numpy as np
+    import torch
+    from PIL import Image
+    from eventgrad_amd.data.datasets import (CIFAR10_CLASSES, JpegDirCifar10,
+                                             build_dataset)
+    from eventgrad_amd.config import DataConfig
+
+    rng = np.random.default_rng(0)
+    for split, n in (("train", 3), ("test", 2)):
+        for j, cls in enumerate(CIFAR10_CLASSES):
+            d = tmp_path / split / cls
+            d.mkdir(parents=True)
+            for i in range(n):
+                # class-coded color so channel ORDER is checkable: class 0
+                # strongly red, class 1 strongly green
+                base = np.zeros((32, 32, 3), np.uint8)
+                base[..., j % 3] = 200
+                base += rng.integers(0, 30, base.shape).astype(np.uint8)
+                Image.fromarray(base).save(d / f"{i:04d}.jpg", quality=95)
+
+    ds = JpegDirCifar10(str(tmp_path), train=True)
+    assert len(ds) == 30
+    x, y = ds[0]
+    assert x.shape == (3, 32, 32) and x.dtype == torch.float32
+    assert y == 0
+    # reference keeps raw 0..255 floats (no normalization)
+    assert x.max() > 150
+    # class 0 images are red-dominant: channel 0 (R) strongest
+    assert x[0].mean() > x[1].mean() and x[0].mean() > x[2].mean()
+    xb, yb = ds.batch(np.array([0, 3, 6]))
+    assert xb.shape == (3, 3, 32, 32)
+    assert yb.tolist() == [0, 1, 2]
+    # class 1 (automobile) green-dominant in the batch
+    assert xb[1, 1].mean() > xb[1, 0].mean()
+
+    # build_dataset auto-detects the layout
+    cfg = DataConfig(dataset="cifar10", data_path=str(tmp_path))
+    d2 = build_dataset(cfg, train=False)
+    assert isinstance(d2, JpegDirCifar10) and len(d2) == 20
