@@ -303,6 +303,166 @@ __global__ __launch_bounds__(TBM * 4) void conv_mm_kernel(
   }
 }
 
+// ---- stride-2 dgrad, parity-decomposed ----------------------------------
+// A stride-2 dgrad through the generic MODE=1 path wastes 3/4 of its MFMA
+// work: for a fixed dx pixel only the (r,s) with matching parity satisfy
+// (h+pad-r) % 2 == 0, so 75% of staged fragments are masked zeros. This
+// kernel runs one launch per parity class (ph,pw) in {0,1}^2 over the
+// quarter-resolution pixel grid h=2*hh+ph, w=2*ww+pw with the reduced
+// filter footprint r = r0+2*ri (ri < Rp), s = s0+2*si — total work is
+// exactly 1/4 of the naive form. ho = hh + cr - ri with cr constant per
+// class, so the cursor stays O(1) add+wrap.
+
+template <int TBM>
+__global__ __launch_bounds__(TBM * 4) void conv_dgrad_p_kernel(
+    const bf16* __restrict__ dy, const bf16* __restrict__ wt,  // CRSK
+    bf16* __restrict__ dx, Geom g, long M, long RED, int ph, int pw,
+    int r0, int s0, int Rp, int Sp, int cr, int cs, int Hh, int Wh) {
+  constexpr int THREADS = TBM * 4;
+  constexpr int BROWS = 512 / THREADS;
+  __shared__ __bf16 smem[TBM * CLDK + CBN * CLDK];
+  __bf16* sA = smem;
+  __bf16* sB = smem + TBM * CLDK;
+
+  const long m0 = (long)blockIdx.x * TBM;
+  const long n0 = (long)blockIdx.y * CBN;   // dx channel tile (C dim)
+  const int t = threadIdx.x;
+  const int lane = t & 63;
+  const int wave = t >> 6;
+  const int wr = wave >> 1, wc = wave & 1;
+  const int lra = t >> 2;
+  const int lka = (t & 3) * 8;
+  const int lrb = t >> 3;
+  const int lkb = (t & 7) * 8;
+
+  // A-row pixel decomposition over the quarter grid
+  const long m = m0 + lra;
+  const bool mvalid = m < M;
+  int pn = 0, phh = 0, pww = 0;
+  if (mvalid) {
+    const int HW = Hh * Wh;
+    pn = (int)(m / HW);
+    int rem = (int)(m % HW);
+    phh = rem / Wh;
+    pww = rem % Wh;
+  }
+  // A reduction cursors over (ri, si, k): inner = K (multiple of 64 here)
+  int acur_k[2], acur_rsi[2];
+#pragma unroll
+  for (int p = 0; p < 2; ++p) {
+    acur_k[p] = lka + 32 * p;   // < 64 <= K
+    acur_rsi[p] = 0;
+  }
+  // B staging cursor for column lkb (k part < 64 <= K)
+  int bcur_k = lkb, bcur_rsi = 0;
+  const long CRSK_row = (long)g.R * g.S * g.K;  // wt row stride (per c)
+
+  __bf16 ra[2][8], rb[BROWS][8];
+
+  auto stage = [&](long k0) {
+#pragma unroll
+    for (int p = 0; p < 2; ++p) {
+      const int ri = acur_rsi[p] / Sp, si = acur_rsi[p] % Sp;
+      const int ho = phh + cr - ri;
+      const int wo = pww + cs - si;
+      const bool valid = mvalid && (unsigned)ho < (unsigned)g.Ho &&
+                         (unsigned)wo < (unsigned)g.Wo;
+      if (valid) {
+        const long off =
+            (((long)pn * g.Ho + ho) * g.Wo + wo) * g.K + acur_k[p];
+        *reinterpret_cast<s16x8*>(ra[p]) =
+            *reinterpret_cast<const s16x8*>(dy + off);
+      } else {
+        zero8(ra[p]);
+      }
+      acur_k[p] += CBK;
+      while (acur_k[p] >= g.K) {
+        acur_k[p] -= g.K;
+        acur_rsi[p] += 1;
+      }
+    }
+    // B: wt[c][r0+2*ri][s0+2*si][k] — vec8 along k
+    {
+      const int ri = bcur_rsi / Sp, si = bcur_rsi % Sp;
+      const long col_off =
+          ((long)(r0 + 2 * ri) * g.S + (s0 + 2 * si)) * g.K + bcur_k;
+#pragma unroll
+      for (int q = 0; q < BROWS; ++q) {
+        const long c = n0 + lrb + q * (THREADS / 8);
+        if (c < g.C && k0 + lkb < RED) {
+          *reinterpret_cast<s16x8*>(rb[q]) =
+              *reinterpret_cast<const s16x8*>(wt + c * CRSK_row + col_off);
+        } else {
+          zero8(rb[q]);
+        }
+      }
+      bcur_k += CBK;
+      while (bcur_k >= g.K) {
+        bcur_k -= g.K;
+        bcur_rsi += 1;
+      }
+    }
+  };
+
+  f32x4 acc[2][2] = {};
+  stage(0);
+
+  for (long k0 = 0; k0 < RED; k0 += CBK) {
+    __syncthreads();
+    *reinterpret_cast<bf16x8*>(&sA[lra * CLDK + lka]) =
+        *reinterpret_cast<bf16x8*>(ra[0]);
+    *reinterpret_cast<bf16x8*>(&sA[lra * CLDK + lka + 32]) =
+        *reinterpret_cast<bf16x8*>(ra[1]);
+#pragma unroll
+    for (int q = 0; q < BROWS; ++q) {
+      *reinterpret_cast<bf16x8*>(
+          &sB[(lrb + q * (THREADS / 8)) * CLDK + lkb]) =
+          *reinterpret_cast<bf16x8*>(rb[q]);
+    }
+    if (k0 + CBK < RED) stage(k0 + CBK);
+    __syncthreads();
+
+    const int ml = lane & 15;
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks) {
+      const int kf = ks * 32 + (lane >> 4) * 8;
+#pragma unroll
+      for (int fi = 0; fi < 2; ++fi) {
+        bf16x8 af = lds8(&sA[(wr * 32 + fi * 16 + ml) * CLDK + kf]);
+#pragma unroll
+        for (int fj = 0; fj < 2; ++fj) {
+          bf16x8 bfr = lds8(&sB[(wc * 32 + fj * 16 + ml) * CLDK + kf]);
+          acc[fi][fj] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af, bfr, acc[fi][fj], 0, 0, 0);
+        }
+      }
+    }
+  }
+
+  // epilogue: scatter the quarter-grid rows back to full-resolution dx
+  const int cn = lane & 15;
+  const int cm = (lane >> 4) * 4;
+  const int HW = Hh * Wh;
+#pragma unroll
+  for (int fj = 0; fj < 2; ++fj) {
+    long nn = n0 + wc * 32 + fj * 16 + cn;
+    if (nn >= g.C) continue;
+#pragma unroll
+    for (int fi = 0; fi < 2; ++fi) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        long mm = m0 + wr * 32 + fi * 16 + cm + r;
+        if (mm >= M) continue;
+        const int n = (int)(mm / HW);
+        const int rem = (int)(mm % HW);
+        const int h = 2 * (rem / Wh) + ph;
+        const int w = 2 * (rem % Wh) + pw;
+        dx[(((long)n * g.H + h) * g.W + w) * g.C + nn] = f2b(acc[fi][fj][r]);
+      }
+    }
+  }
+}
+
 // ---- wgrad: dw[k, rsc] = sum_np dy[np,k] * im2col[np,rsc]  (TN) ----------
 // 64x64 tile over (k, rsc), WBK=64 deep in np. Both operands need their
 // MFMA fragments transposed relative to the global [np, inner] layout; the
@@ -749,9 +909,50 @@ torch::Tensor conv2d_dgrad(torch::Tensor dy, torch::Tensor wt, long stride,
   TORCH_CHECK((int)dy.size(3) == K);
   g.H = (int)H; g.W = (int)W; g.C = C; g.K = K; g.R = R; g.S = S;
   g.stride = (int)stride; g.pad = (int)pad;
-  auto dx = torch::empty({g.N, g.H, g.W, g.C}, dy.options());
   long M = (long)g.N * g.H * g.W;
   long RED = (long)R * S * K;
+  // stride-2 parity decomposition: 4 quarter-grid launches at 1/4 the
+  // total work (the naive MODE=1 path masks 75% of fragments to zero)
+  if (g.stride == 2 && g.K % 64 == 0) {
+    bool any_empty = false;
+    for (int p = 0; p < 2 && !any_empty; ++p) {
+      int r0 = (p + g.pad) & 1;  // same formula for the s dimension
+      if ((R - r0 + 1) / 2 <= 0 || (S - r0 + 1) / 2 <= 0)
+        any_empty = true;        // e.g. the 1x1 downsampler's odd classes
+    }
+    auto dx = any_empty
+        ? torch::zeros({g.N, g.H, g.W, g.C}, dy.options())
+        : torch::empty({g.N, g.H, g.W, g.C}, dy.options());
+    for (int ph = 0; ph < 2; ++ph) {
+      if (ph >= g.H) continue;
+      const int r0 = (ph + g.pad) & 1;
+      const int Rp = (R - r0 + 1) / 2;
+      if (Rp <= 0) continue;
+      const int cr = (ph + g.pad - r0) >> 1;
+      const int Hh = (g.H - ph + 1) >> 1;
+      for (int pw = 0; pw < 2; ++pw) {
+        if (pw >= g.W) continue;
+        const int s0 = (pw + g.pad) & 1;
+        const int Sp = (S - s0 + 1) / 2;
+        if (Sp <= 0) continue;
+        const int cs = (pw + g.pad - s0) >> 1;
+        const int Wh = (g.W - pw + 1) >> 1;
+        const long Mp = (long)g.N * Hh * Wh;
+        const long REDp = (long)Rp * Sp * K;
+        long blocks128 = (long)ceil_div(Mp, 128) * ceil_div(C, conv::CBN);
+        long tbm = blocks128 < 384 ? 64 : 128;
+        dim3 grid(ceil_div(Mp, tbm), ceil_div(C, conv::CBN));
+        auto* fn = tbm == 64 ? conv::conv_dgrad_p_kernel<64>
+                             : conv::conv_dgrad_p_kernel<128>;
+        fn<<<grid, (unsigned)(tbm * 4), 0, cur_stream()>>>(
+            (const bf16*)dy.data_ptr(), (const bf16*)wt.data_ptr(),
+            (bf16*)dx.data_ptr(), g, Mp, REDp, ph, pw, r0, s0, Rp, Sp,
+            cr, cs, Hh, Wh);
+      }
+    }
+    return dx;
+  }
+  auto dx = torch::empty({g.N, g.H, g.W, g.C}, dy.options());
   bool fast = (g.K % 8 == 0) && (g.stride <= 2);
   long blocks128 = (long)ceil_div(M, 128) * ceil_div(C, conv::CBN);
   bool narrow = blocks128 < 384;
