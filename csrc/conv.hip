@@ -45,6 +45,19 @@ __device__ __forceinline__ void zero8(__bf16* dst) {
   for (int j = 0; j < 8; ++j) dst[j] = (__bf16)0.f;
 }
 
+// XCD-aware tile remap (bijective; guide §6). The dispatcher places linear
+// block b on XCD b % 8, so consecutive blockIdx.x tiles — whose im2col
+// windows overlap in x — land on DIFFERENT 4 MiB L2s and re-read HBM/L3.
+// Remap so each XCD owns a CONTIGUOUS M-tile range across all N-tiles:
+// its L2 then serves the R*S-fold A re-reads.
+__device__ __forceinline__ long xcd_tile_remap() {
+  const long T = (long)gridDim.x * gridDim.y;
+  const long orig = (long)blockIdx.x + (long)gridDim.x * blockIdx.y;
+  const long q = T >> 3, r = T & 7;
+  const long xcd = orig & 7, idx = orig >> 3;
+  return (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
+}
+
 // ---- generic (slow) stagers: per-element guarded loads -------------------
 
 __device__ __forceinline__ void stage_im2col_slow(const bf16* __restrict__ x,
@@ -136,8 +149,9 @@ __global__ __launch_bounds__(TBM * 4) void conv_mm_kernel(
     s_sq[threadIdx.x] = 0.f;
   }
 
-  const long m0 = (long)blockIdx.x * TBM;
-  const long n0 = (long)blockIdx.y * CBN;
+  const long wg = xcd_tile_remap();  // m-major: contiguous M per XCD
+  const long m0 = (wg / gridDim.y) * TBM;
+  const long n0 = (wg % gridDim.y) * CBN;
   const int t = threadIdx.x;
   const int lane = t & 63;
   const int wave = t >> 6;
@@ -324,8 +338,9 @@ __global__ __launch_bounds__(TBM * 4) void conv_dgrad_p_kernel(
   __bf16* sA = smem;
   __bf16* sB = smem + TBM * CLDK;
 
-  const long m0 = (long)blockIdx.x * TBM;
-  const long n0 = (long)blockIdx.y * CBN;   // dx channel tile (C dim)
+  const long wg = xcd_tile_remap();
+  const long m0 = (wg / gridDim.y) * TBM;
+  const long n0 = (wg % gridDim.y) * CBN;   // dx channel tile (C dim)
   const int t = threadIdx.x;
   const int lane = t & 63;
   const int wave = t >> 6;
