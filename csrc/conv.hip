@@ -58,6 +58,19 @@ __device__ __forceinline__ long xcd_tile_remap() {
   return (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
 }
 
+// 3D variant for the wgrad grid (k-tiles × rsc-tiles × np-slices): all the
+// (k, rsc) tiles of one np-slice read the SAME x/dy rows (the rsc tiles
+// re-read x R*S-fold), so each XCD gets contiguous z-major chunks.
+__device__ __forceinline__ long xcd_tile_remap3() {
+  const long T = (long)gridDim.x * gridDim.y * gridDim.z;
+  const long orig = (long)blockIdx.x +
+                    (long)gridDim.x * (blockIdx.y +
+                                       (long)gridDim.y * blockIdx.z);
+  const long q = T >> 3, r = T & 7;
+  const long xcd = orig & 7, idx = orig >> 3;
+  return (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
+}
+
 // ---- generic (slow) stagers: per-element guarded loads -------------------
 
 __device__ __forceinline__ void stage_im2col_slow(const bf16* __restrict__ x,
@@ -327,20 +340,36 @@ __global__ __launch_bounds__(TBM * 4) void conv_mm_kernel(
 // exactly 1/4 of the naive form. ho = hh + cr - ri with cr constant per
 // class, so the cursor stays O(1) add+wrap.
 
+// per-parity-class geometry, all four classes dispatched in ONE launch
+// (separate per-class launches left the deep stages' quarter grids at
+// <1 block/CU each)
+struct PClass {
+  int ph, pw, r0, s0, Rp, Sp, cr, cs, Hh, Wh;
+  long M, RED;
+};
+struct PClasses {
+  PClass c[4];
+};
+
 template <int TBM>
 __global__ __launch_bounds__(TBM * 4) void conv_dgrad_p_kernel(
     const bf16* __restrict__ dy, const bf16* __restrict__ wt,  // CRSK
-    bf16* __restrict__ dx, Geom g, long M, long RED, int ph, int pw,
-    int r0, int s0, int Rp, int Sp, int cr, int cs, int Hh, int Wh) {
+    bf16* __restrict__ dx, Geom g, PClasses pcs) {
   constexpr int THREADS = TBM * 4;
   constexpr int BROWS = 512 / THREADS;
   __shared__ __bf16 smem[TBM * CLDK + CBN * CLDK];
   __bf16* sA = smem;
   __bf16* sB = smem + TBM * CLDK;
 
-  const long wg = xcd_tile_remap();
+  const PClass P = pcs.c[blockIdx.z];
+  const int ph = P.ph, pw = P.pw, r0 = P.r0, s0 = P.s0;
+  const int Rp = P.Rp, Sp = P.Sp, cr = P.cr, cs = P.cs;
+  const int Hh = P.Hh, Wh = P.Wh;
+  const long M = P.M, RED = P.RED;
+  const long wg = xcd_tile_remap();  // 2D remap within this class slice
   const long m0 = (wg / gridDim.y) * TBM;
   const long n0 = (wg % gridDim.y) * CBN;   // dx channel tile (C dim)
+  if (m0 >= M) return;                       // class smaller than grid.x
   const int t = threadIdx.x;
   const int lane = t & 63;
   const int wave = t >> 6;
@@ -532,9 +561,13 @@ __global__ __launch_bounds__(256) void conv_wgrad_fast_kernel(
   __shared__ __bf16 sA[4 * WIMG];
   __shared__ __bf16 sB[4 * WIMG];
 
-  const long k0c = (long)blockIdx.x * WBM;
-  const long n0 = (long)blockIdx.y * WBN;
-  const long np0 = (long)blockIdx.z * npslice;
+  const long wgid = xcd_tile_remap3();  // z-major chunks per XCD
+  const long bx = wgid % gridDim.x;
+  const long by = (wgid / gridDim.x) % gridDim.y;
+  const long bz = wgid / ((long)gridDim.x * gridDim.y);
+  const long k0c = bx * WBM;
+  const long n0 = by * WBN;
+  const long np0 = bz * npslice;
   const long np1 = min(np0 + npslice, NP);
   const long RED = (long)g.R * g.S * g.C;
 
@@ -679,9 +712,10 @@ __global__ __launch_bounds__(256) void conv_wgrad_kernel(
   __shared__ __bf16 sA[WBM * WLDN];
   __shared__ __bf16 sB[WBN * WLDN];
 
-  const long k0c = (long)blockIdx.x * WBM;   // out-channel tile
-  const long n0 = (long)blockIdx.y * WBN;    // rsc tile
-  const long np0 = (long)blockIdx.z * npslice;
+  const long wgid = xcd_tile_remap3();       // z-major chunks per XCD
+  const long k0c = (wgid % gridDim.x) * WBM;           // out-channel tile
+  const long n0 = ((wgid / gridDim.x) % gridDim.y) * WBN;  // rsc tile
+  const long np0 = (wgid / ((long)gridDim.x * gridDim.y)) * npslice;
   const long np1 = min(np0 + npslice, NP);
   const long RED = (long)g.R * g.S * g.C;
 
@@ -929,41 +963,45 @@ torch::Tensor conv2d_dgrad(torch::Tensor dy, torch::Tensor wt, long stride,
   // stride-2 parity decomposition: 4 quarter-grid launches at 1/4 the
   // total work (the naive MODE=1 path masks 75% of fragments to zero)
   if (g.stride == 2 && g.K % 64 == 0) {
+    conv::PClasses pcs{};
+    int nc = 0;
     bool any_empty = false;
-    for (int p = 0; p < 2 && !any_empty; ++p) {
-      int r0 = (p + g.pad) & 1;  // same formula for the s dimension
-      if ((R - r0 + 1) / 2 <= 0 || (S - r0 + 1) / 2 <= 0)
-        any_empty = true;        // e.g. the 1x1 downsampler's odd classes
+    long maxM = 0;
+    for (int ph = 0; ph < 2; ++ph) {
+      const int r0 = (ph + g.pad) & 1;
+      const int Rp = ((int)R - r0 + 1) / 2;
+      for (int pw = 0; pw < 2; ++pw) {
+        const int s0 = (pw + g.pad) & 1;
+        const int Sp = ((int)S - s0 + 1) / 2;
+        if (Rp <= 0 || Sp <= 0 || ph >= g.H || pw >= g.W) {
+          any_empty = true;  // e.g. the 1x1 downsampler's odd classes
+          continue;
+        }
+        conv::PClass& P = pcs.c[nc++];
+        P.ph = ph; P.pw = pw; P.r0 = r0; P.s0 = s0;
+        P.Rp = Rp; P.Sp = Sp;
+        P.cr = (ph + g.pad - r0) >> 1;
+        P.cs = (pw + g.pad - s0) >> 1;
+        P.Hh = (g.H - ph + 1) >> 1;
+        P.Wh = (g.W - pw + 1) >> 1;
+        P.M = (long)g.N * P.Hh * P.Wh;
+        P.RED = (long)Rp * Sp * K;
+        maxM = std::max(maxM, P.M);
+      }
     }
     auto dx = any_empty
         ? torch::zeros({g.N, g.H, g.W, g.C}, dy.options())
         : torch::empty({g.N, g.H, g.W, g.C}, dy.options());
-    for (int ph = 0; ph < 2; ++ph) {
-      if (ph >= g.H) continue;
-      const int r0 = (ph + g.pad) & 1;
-      const int Rp = (R - r0 + 1) / 2;
-      if (Rp <= 0) continue;
-      const int cr = (ph + g.pad - r0) >> 1;
-      const int Hh = (g.H - ph + 1) >> 1;
-      for (int pw = 0; pw < 2; ++pw) {
-        if (pw >= g.W) continue;
-        const int s0 = (pw + g.pad) & 1;
-        const int Sp = (S - s0 + 1) / 2;
-        if (Sp <= 0) continue;
-        const int cs = (pw + g.pad - s0) >> 1;
-        const int Wh = (g.W - pw + 1) >> 1;
-        const long Mp = (long)g.N * Hh * Wh;
-        const long REDp = (long)Rp * Sp * K;
-        long blocks128 = (long)ceil_div(Mp, 128) * ceil_div(C, conv::CBN);
-        long tbm = blocks128 < 384 ? 64 : 128;
-        dim3 grid(ceil_div(Mp, tbm), ceil_div(C, conv::CBN));
-        auto* fn = tbm == 64 ? conv::conv_dgrad_p_kernel<64>
-                             : conv::conv_dgrad_p_kernel<128>;
-        fn<<<grid, (unsigned)(tbm * 4), 0, cur_stream()>>>(
-            (const bf16*)dy.data_ptr(), (const bf16*)wt.data_ptr(),
-            (bf16*)dx.data_ptr(), g, Mp, REDp, ph, pw, r0, s0, Rp, Sp,
-            cr, cs, Hh, Wh);
-      }
+    if (nc > 0) {
+      long blocks128 =
+          (long)ceil_div(maxM, 128) * ceil_div(C, conv::CBN) * nc;
+      long tbm = blocks128 < 384 ? 64 : 128;
+      dim3 grid(ceil_div(maxM, tbm), ceil_div(C, conv::CBN), (unsigned)nc);
+      auto* fn = tbm == 64 ? conv::conv_dgrad_p_kernel<64>
+                           : conv::conv_dgrad_p_kernel<128>;
+      fn<<<grid, (unsigned)(tbm * 4), 0, cur_stream()>>>(
+          (const bf16*)dy.data_ptr(), (const bf16*)wt.data_ptr(),
+          (bf16*)dx.data_ptr(), g, pcs);
     }
     return dx;
   }
