@@ -883,6 +883,17 @@ torch::Tensor tr16_probe(long mode) {
   return out;
 }
 
+// zsplit target: each dw element takes `zsplit` fp32 atomic hits, so the
+// block count trades atomic contention against fill/loop length
+// (EG_WGRAD_TB overrides for tuning sweeps).
+static long wgrad_target_blocks() {
+  static long v = [] {
+    const char* e = getenv("EG_WGRAD_TB");
+    return e ? atol(e) : 2048L;
+  }();
+  return v;
+}
+
 static conv::Geom make_geom(const torch::Tensor& x, int K, int R, int S,
                             long stride, long pad) {
   conv::Geom g;
@@ -1031,7 +1042,7 @@ torch::Tensor conv2d_wgrad(torch::Tensor x, torch::Tensor dy, long R, long S,
   long NP = (long)g.N * g.Ho * g.Wo;
   long RED = (long)R * S * g.C;
   // split the NP reduction across blocks for parallelism; fp32 atomics
-  long target_blocks = 2048;
+  long target_blocks = wgrad_target_blocks();
   long tiles = (long)ceil_div(g.K, conv::WBM) * ceil_div(RED, conv::WBN);
   long zsplit =
       std::max(1L, std::min(512L, target_blocks / std::max(tiles, 1L)));
@@ -1120,7 +1131,7 @@ void conv2d_wgrad_into(torch::Tensor x, torch::Tensor dy, torch::Tensor dw,
   long total = (long)g.K * RED;
   TORCH_CHECK(dw.numel() == total);
   long NP = (long)g.N * g.Ho * g.Wo;
-  long target_blocks = 2048;
+  long target_blocks = wgrad_target_blocks();
   long tiles = (long)ceil_div(g.K, conv::WBM) * ceil_div(RED, conv::WBN);
   long zsplit =
       std::max(1L, std::min(512L, target_blocks / std::max(tiles, 1L)));
