@@ -152,10 +152,13 @@ __global__ __launch_bounds__(TBM * 4) void conv_mm_kernel(
   // makes hipcc over-synchronize the k-loop — measured +14 waitcnts and an
   // extra barrier per step with separate stats arrays). Stats accumulators
   // live in a float-aliased tail of the same array.
-  __shared__ __bf16 smem[TBM * CLDK + CBN * CLDK + (STATS ? 2 * CBN * 2 : 0)];
-  __bf16* sA = smem;
-  __bf16* sB = smem + TBM * CLDK;
-  float* s_sum = reinterpret_cast<float*>(smem + TBM * CLDK + CBN * CLDK);
+  // double-buffered LDS for the TBM=64 (deep, issue-bound) shapes: one
+  // barrier per k-tile instead of two. TBM=128 keeps a single buffer —
+  // doubling its 27.6 KB LDS halves occupancy and measured net-slower.
+  constexpr bool DB = (TBM == 64);
+  constexpr int BUFSZ = TBM * CLDK + CBN * CLDK;
+  __shared__ __bf16 smem[(DB ? 2 : 1) * BUFSZ + (STATS ? 2 * CBN * 2 : 0)];
+  float* s_sum = reinterpret_cast<float*>(smem + (DB ? 2 : 1) * BUFSZ);
   float* s_sq = s_sum + CBN;
   if (STATS && threadIdx.x < CBN) {
     s_sum[threadIdx.x] = 0.f;
@@ -248,23 +251,43 @@ __global__ __launch_bounds__(TBM * 4) void conv_mm_kernel(
     }
   };
 
-  f32x4 acc[2][2] = {};
-  stage(0);
-
-  for (long k0 = 0; k0 < RED; k0 += CBK) {
-    __syncthreads();  // prior tile's fragment reads done; LDS reusable
-    *reinterpret_cast<bf16x8*>(&sA[lra * CLDK + lka]) =
+  auto write_lds = [&](int pb) {
+    __bf16* wA = smem + pb * BUFSZ;
+    __bf16* wB = wA + TBM * CLDK;
+    *reinterpret_cast<bf16x8*>(&wA[lra * CLDK + lka]) =
         *reinterpret_cast<bf16x8*>(ra[0]);
-    *reinterpret_cast<bf16x8*>(&sA[lra * CLDK + lka + 32]) =
+    *reinterpret_cast<bf16x8*>(&wA[lra * CLDK + lka + 32]) =
         *reinterpret_cast<bf16x8*>(ra[1]);
 #pragma unroll
     for (int q = 0; q < BROWS; ++q) {
       *reinterpret_cast<bf16x8*>(
-          &sB[(lrb + q * (THREADS / 8)) * CLDK + lkb]) =
+          &wB[(lrb + q * (THREADS / 8)) * CLDK + lkb]) =
           *reinterpret_cast<bf16x8*>(rb[q]);
     }
-    if (k0 + CBK < RED) stage(k0 + CBK);  // overlap loads with MFMA below
-    __syncthreads();
+  };
+
+  f32x4 acc[2][2] = {};
+  stage(0);
+  if (DB) {
+    write_lds(0);                     // no readers yet: no barrier needed
+    if (CBK < RED) stage(CBK);
+  }
+  int pb = 0;
+
+  for (long k0 = 0; k0 < RED; k0 += CBK) {
+    __syncthreads();  // LDS safe: prior tile's reads (and writes) done
+    if (DB) {
+      if (k0 + CBK < RED) {
+        write_lds(pb ^ 1);            // tile k0+CBK from staged regs
+        if (k0 + 2 * CBK < RED) stage(k0 + 2 * CBK);  // loads overlap MFMA
+      }
+    } else {
+      write_lds(0);
+      if (k0 + CBK < RED) stage(k0 + CBK);
+      __syncthreads();
+    }
+    const __bf16* sA = smem + pb * BUFSZ;
+    const __bf16* sB = sA + TBM * CLDK;
 
     const int ml = lane & 15;
 #pragma unroll
@@ -281,6 +304,7 @@ __global__ __launch_bounds__(TBM * 4) void conv_mm_kernel(
         }
       }
     }
+    if (DB) pb ^= 1;
   }
 
   const int cn = lane & 15;
@@ -357,9 +381,9 @@ __global__ __launch_bounds__(TBM * 4) void conv_dgrad_p_kernel(
     bf16* __restrict__ dx, Geom g, PClasses pcs) {
   constexpr int THREADS = TBM * 4;
   constexpr int BROWS = 512 / THREADS;
-  __shared__ __bf16 smem[TBM * CLDK + CBN * CLDK];
-  __bf16* sA = smem;
-  __bf16* sB = smem + TBM * CLDK;
+  constexpr bool DB = (TBM == 64);   // see conv_mm_kernel note
+  constexpr int BUFSZ = TBM * CLDK + CBN * CLDK;
+  __shared__ __bf16 smem[(DB ? 2 : 1) * BUFSZ];
 
   const PClass P = pcs.c[blockIdx.z];
   const int ph = P.ph, pw = P.pw, r0 = P.r0, s0 = P.s0;
@@ -448,23 +472,43 @@ __global__ __launch_bounds__(TBM * 4) void conv_dgrad_p_kernel(
     }
   };
 
-  f32x4 acc[2][2] = {};
-  stage(0);
-
-  for (long k0 = 0; k0 < RED; k0 += CBK) {
-    __syncthreads();
-    *reinterpret_cast<bf16x8*>(&sA[lra * CLDK + lka]) =
+  auto write_lds = [&](int pbuf) {
+    __bf16* wA = smem + pbuf * BUFSZ;
+    __bf16* wB = wA + TBM * CLDK;
+    *reinterpret_cast<bf16x8*>(&wA[lra * CLDK + lka]) =
         *reinterpret_cast<bf16x8*>(ra[0]);
-    *reinterpret_cast<bf16x8*>(&sA[lra * CLDK + lka + 32]) =
+    *reinterpret_cast<bf16x8*>(&wA[lra * CLDK + lka + 32]) =
         *reinterpret_cast<bf16x8*>(ra[1]);
 #pragma unroll
     for (int q = 0; q < BROWS; ++q) {
       *reinterpret_cast<bf16x8*>(
-          &sB[(lrb + q * (THREADS / 8)) * CLDK + lkb]) =
+          &wB[(lrb + q * (THREADS / 8)) * CLDK + lkb]) =
           *reinterpret_cast<bf16x8*>(rb[q]);
     }
-    if (k0 + CBK < RED) stage(k0 + CBK);
+  };
+
+  f32x4 acc[2][2] = {};
+  stage(0);
+  if (DB) {
+    write_lds(0);
+    if (CBK < RED) stage(CBK);
+  }
+  int pb = 0;
+
+  for (long k0 = 0; k0 < RED; k0 += CBK) {
     __syncthreads();
+    if (DB) {
+      if (k0 + CBK < RED) {
+        write_lds(pb ^ 1);
+        if (k0 + 2 * CBK < RED) stage(k0 + 2 * CBK);
+      }
+    } else {
+      write_lds(0);
+      if (k0 + CBK < RED) stage(k0 + CBK);
+      __syncthreads();
+    }
+    const __bf16* sA = smem + pb * BUFSZ;
+    const __bf16* sB = sA + TBM * CLDK;
 
     const int ml = lane & 15;
 #pragma unroll
@@ -481,6 +525,7 @@ __global__ __launch_bounds__(TBM * 4) void conv_dgrad_p_kernel(
         }
       }
     }
+    if (DB) pb ^= 1;
   }
 
   // epilogue: scatter the quarter-grid rows back to full-resolution dx
