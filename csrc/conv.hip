@@ -189,15 +189,71 @@ __global__ __launch_bounds__(TBM * 4) void conv_mm_kernel(
     ph = rem / WW;
     pw = rem % WW;
   }
-  // reduction cursors for the two staged vec8 positions (lka, lka+32).
-  // inner = c (MODE0) or k (MODE1); rs = r*S + s.
+  // FAST staging state (PMC: recomputed address chains made these kernels
+  // VALU-bound at 14-18 VALU instructions per MFMA). Per thread:
+  //  - a validity BITMASK over the R*S filter taps (computed once; the
+  //    per-tap h/w range checks collapse to one shift+and);
+  //  - an incremental source offset with CONSTANT wrap corrections
+  //    (MODE 0: (s,c) is contiguous in NHWC x, so within a filter row the
+  //    offset just advances by CBK; crossing to the next r adds
+  //    (W-S)*C. MODE 1 (stride 1): s-wrap -2K, r-wrap K*(S-Wo-2)).
+  // inner = c (MODE0) or k (MODE1); requires R*S <= 32 (host-gated).
   const int INNER = MODE == 0 ? g.C : g.K;
-  int cur_i[2], cur_rs[2];
+  unsigned long long vmask = 0;
+  long aoff[2];
+  int cur_i[2], rsn[2], scnt[2];
+  long swrap_fix = 0, rwrap_fix = 0;
+  if (FAST) {
+    if (MODE == 0) {
+      const int hb = ph * g.stride - g.pad;   // h = hb + r
+      const int wb = pw * g.stride - g.pad;   // w = wb + s
+      for (int r = 0; r < g.R; ++r)
+        for (int s2 = 0; s2 < g.S; ++s2)
+          if ((unsigned)(hb + r) < (unsigned)g.H &&
+              (unsigned)(wb + s2) < (unsigned)g.W)
+            vmask |= 1ull << (r * g.S + s2);
+      swrap_fix = 0;
+      rwrap_fix = (long)(g.W - g.S) * g.C;
+      for (int p = 0; p < 2; ++p) {
+        const int red = lka + 32 * p;
+        cur_i[p] = red % INNER;
+        rsn[p] = red / INNER;
+        const int r = rsn[p] / g.S, s2 = rsn[p] % g.S;
+        scnt[p] = s2;
+        aoff[p] = (((long)pn * g.H + hb + r) * g.W + wb + s2) * g.C +
+                  cur_i[p];
+      }
+    } else {
+      const int hob = ph + g.pad;             // ho = hob - r (stride 1)
+      const int wob = pw + g.pad;
+      for (int r = 0; r < g.R; ++r)
+        for (int s2 = 0; s2 < g.S; ++s2)
+          if ((unsigned)(hob - r) < (unsigned)g.Ho &&
+              (unsigned)(wob - s2) < (unsigned)g.Wo)
+            vmask |= 1ull << (r * g.S + s2);
+      swrap_fix = -2L * g.K;
+      rwrap_fix = (long)g.K * (g.S - 2 - g.Wo);
+      for (int p = 0; p < 2; ++p) {
+        const int red = lka + 32 * p;
+        cur_i[p] = red % INNER;
+        rsn[p] = red / INNER;
+        const int r = rsn[p] / g.S, s2 = rsn[p] % g.S;
+        scnt[p] = s2;
+        aoff[p] = (((long)pn * g.Ho + hob - r) * g.Wo + (wob - s2)) * g.K +
+                  cur_i[p];
+      }
+    }
+  }
+
+  // B rows are fixed per thread (plain [N, RED] weight matrix): hoist the
+  // guards and advance a raw pointer by CBK per tile
+  const bf16* bptr[BROWS];
+  bool bok[BROWS];
 #pragma unroll
-  for (int p = 0; p < 2; ++p) {
-    int red = lka + 32 * p;
-    cur_i[p] = FAST ? red % INNER : red;  // slow path keeps raw red
-    cur_rs[p] = FAST ? red / INNER : 0;
+  for (int q = 0; q < BROWS; ++q) {
+    const long rrow = n0 + lrb + q * (THREADS / 8);
+    bok[q] = rrow < N;
+    bptr[q] = B + rrow * RED + lkb;
   }
 
   __bf16 ra[2][8], rb[BROWS][8];
@@ -206,35 +262,24 @@ __global__ __launch_bounds__(TBM * 4) void conv_mm_kernel(
 #pragma unroll
     for (int p = 0; p < 2; ++p) {
       if (FAST) {
-        const int r = cur_rs[p] / g.S, s = cur_rs[p] % g.S;
-        bool valid;
-        long off;
-        if (MODE == 0) {
-          int h = ph * g.stride - g.pad + r;
-          int w = pw * g.stride - g.pad + s;
-          valid = mvalid && (unsigned)h < (unsigned)g.H &&
-                  (unsigned)w < (unsigned)g.W;
-          off = (((long)pn * g.H + h) * g.W + w) * g.C + cur_i[p];
-        } else {
-          int hq = ph + g.pad - r, wq = pw + g.pad - s;
-          bool div_ok = (g.stride == 1) || (((hq | wq) & 1) == 0);
-          int ho = g.stride == 2 ? (hq >> 1) : hq;
-          int wo = g.stride == 2 ? (wq >> 1) : wq;
-          valid = mvalid && hq >= 0 && wq >= 0 && div_ok && ho < g.Ho &&
-                  wo < g.Wo;
-          off = (((long)pn * g.Ho + ho) * g.Wo + wo) * g.K + cur_i[p];
-        }
+        const bool valid = mvalid && ((vmask >> rsn[p]) & 1ull);
         if (valid) {
           *reinterpret_cast<s16x8*>(ra[p]) =
-              *reinterpret_cast<const s16x8*>(Asrc + off);
+              *reinterpret_cast<const s16x8*>(Asrc + aoff[p]);
         } else {
           zero8(ra[p]);
         }
-        // advance cursor by CBK elements along the inner dim
+        aoff[p] += CBK;
         cur_i[p] += CBK;
         while (cur_i[p] >= INNER) {
           cur_i[p] -= INNER;
-          cur_rs[p] += 1;
+          rsn[p] += 1;
+          if (++scnt[p] == g.S) {
+            scnt[p] = 0;
+            aoff[p] += rwrap_fix;
+          } else {
+            aoff[p] += swrap_fix;
+          }
         }
       } else {
         int red = (int)k0 + lka + 32 * p;
@@ -245,9 +290,22 @@ __global__ __launch_bounds__(TBM * 4) void conv_mm_kernel(
         }
       }
     }
+    if (FAST) {
 #pragma unroll
-    for (int q = 0; q < BROWS; ++q) {
-      row8(B, N, RED, n0 + lrb + q * (THREADS / 8), k0 + lkb, rb[q]);
+      for (int q = 0; q < BROWS; ++q) {
+        if (bok[q] && k0 + lkb < RED) {  // RED % 8 == 0 in FAST mode
+          *reinterpret_cast<s16x8*>(rb[q]) =
+              *reinterpret_cast<const s16x8*>(bptr[q]);
+        } else {
+          zero8(rb[q]);
+        }
+        bptr[q] += CBK;
+      }
+    } else {
+#pragma unroll
+      for (int q = 0; q < BROWS; ++q) {
+        row8(B, N, RED, n0 + lrb + q * (THREADS / 8), k0 + lkb, rb[q]);
+      }
     }
   };
 
@@ -966,7 +1024,7 @@ torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor bias,
   long M = (long)g.N * g.Ho * g.Wo;
   long RED = (long)g.R * g.S * g.C;
   bool has_bias = bias.numel() > 0;
-  bool fast = (g.C % 8 == 0);
+  bool fast = (g.C % 8 == 0) && (g.R * g.S <= 32);
   // shape-adaptive tile: 128-row tiles starve the 256-CU chip on the deep
   // stages (M drops to 4096); switch to 64-row tiles below ~384 blocks.
   long blocks128 = (long)ceil_div(M, 128) * ceil_div(g.K, conv::CBN);
@@ -1062,7 +1120,9 @@ torch::Tensor conv2d_dgrad(torch::Tensor dy, torch::Tensor wt, long stride,
     return dx;
   }
   auto dx = torch::empty({g.N, g.H, g.W, g.C}, dy.options());
-  bool fast = (g.K % 8 == 0) && (g.stride <= 2);
+  // FAST MODE1 handles stride 1 only (stride 2 goes to the parity path
+  // when K % 64 == 0, else the generic slow path)
+  bool fast = (g.K % 8 == 0) && (g.stride == 1) && (g.R * g.S <= 32);
   long blocks128 = (long)ceil_div(M, 128) * ceil_div(C, conv::CBN);
   bool narrow = blocks128 < 384;
   long tbm = narrow ? 64 : 128;
