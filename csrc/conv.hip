@@ -472,60 +472,91 @@ __global__ __launch_bounds__(TBM * 4) void conv_dgrad_p_kernel(
     phh = rem / Wh;
     pww = rem % Wh;
   }
-  // A reduction cursors over (ri, si, k): inner = K (multiple of 64 here)
-  int acur_k[2], acur_rsi[2];
+  // staging state: validity bitmask over the (ri, si) taps (<= 4 bits) +
+  // incremental offsets with constant wrap fixes — the div/mod form
+  // measured 38 VALU instructions per MFMA (runtime int division)
+  unsigned vmask = 0;
+  for (int ri = 0; ri < Rp; ++ri)
+    for (int si = 0; si < Sp; ++si)
+      if ((unsigned)(phh + cr - ri) < (unsigned)g.Ho &&
+          (unsigned)(pww + cs - si) < (unsigned)g.Wo)
+        vmask |= 1u << (ri * Sp + si);
+  const long a_swrap = -2L * g.K;                       // si+1: wo -= 1
+  const long a_rwrap = (long)g.K * (Sp - 2 - g.Wo);     // ri+1: ho -= 1
+  long aoff[2];
+  int acur_k[2], arsn[2], asct[2];
 #pragma unroll
   for (int p = 0; p < 2; ++p) {
     acur_k[p] = lka + 32 * p;   // < 64 <= K
-    acur_rsi[p] = 0;
+    arsn[p] = 0;
+    asct[p] = 0;
+    aoff[p] = (((long)pn * g.Ho + phh + cr) * g.Wo + pww + cs) * g.K +
+              acur_k[p];
   }
-  // B staging cursor for column lkb (k part < 64 <= K)
-  int bcur_k = lkb, bcur_rsi = 0;
+  // B columns: wt[c][r0+2*ri][s0+2*si][lkb + k]; per-row base pointers
+  // hoisted, one shared column offset advanced with constant wrap fixes
   const long CRSK_row = (long)g.R * g.S * g.K;  // wt row stride (per c)
+  // si+1 advances s by 2 (+2K) while k resets (-K): net +K
+  const long b_sw = (long)g.K;
+  // ri+1: r += 2 (+2*S*K), si resets (-(Sp-1)*2K), k resets (-K)
+  const long b_rw = 2L * g.S * g.K - 2L * g.K * (Sp - 1) - (long)g.K;
+  long bcol = ((long)r0 * g.S + s0) * g.K + lkb;
+  int bcur_k = lkb, bsct = 0;
+  const bf16* bbase[BROWS];
+  bool bok[BROWS];
+#pragma unroll
+  for (int q = 0; q < BROWS; ++q) {
+    const long c = n0 + lrb + q * (THREADS / 8);
+    bok[q] = c < g.C;
+    bbase[q] = wt + c * CRSK_row;
+  }
 
   __bf16 ra[2][8], rb[BROWS][8];
 
   auto stage = [&](long k0) {
 #pragma unroll
     for (int p = 0; p < 2; ++p) {
-      const int ri = acur_rsi[p] / Sp, si = acur_rsi[p] % Sp;
-      const int ho = phh + cr - ri;
-      const int wo = pww + cs - si;
-      const bool valid = mvalid && (unsigned)ho < (unsigned)g.Ho &&
-                         (unsigned)wo < (unsigned)g.Wo;
+      const bool valid = mvalid && arsn[p] < 32 && ((vmask >> arsn[p]) & 1u);
       if (valid) {
-        const long off =
-            (((long)pn * g.Ho + ho) * g.Wo + wo) * g.K + acur_k[p];
         *reinterpret_cast<s16x8*>(ra[p]) =
-            *reinterpret_cast<const s16x8*>(dy + off);
+            *reinterpret_cast<const s16x8*>(dy + aoff[p]);
       } else {
         zero8(ra[p]);
       }
+      aoff[p] += CBK;
       acur_k[p] += CBK;
       while (acur_k[p] >= g.K) {
         acur_k[p] -= g.K;
-        acur_rsi[p] += 1;
+        arsn[p] += 1;
+        if (++asct[p] == Sp) {
+          asct[p] = 0;
+          aoff[p] += a_rwrap;
+        } else {
+          aoff[p] += a_swrap;
+        }
       }
     }
-    // B: wt[c][r0+2*ri][s0+2*si][k] — vec8 along k
+    // B: vec8 along k at the shared strided column offset
     {
-      const int ri = bcur_rsi / Sp, si = bcur_rsi % Sp;
-      const long col_off =
-          ((long)(r0 + 2 * ri) * g.S + (s0 + 2 * si)) * g.K + bcur_k;
 #pragma unroll
       for (int q = 0; q < BROWS; ++q) {
-        const long c = n0 + lrb + q * (THREADS / 8);
-        if (c < g.C && k0 + lkb < RED) {
+        if (bok[q] && k0 + lkb < RED) {
           *reinterpret_cast<s16x8*>(rb[q]) =
-              *reinterpret_cast<const s16x8*>(wt + c * CRSK_row + col_off);
+              *reinterpret_cast<const s16x8*>(bbase[q] + bcol);
         } else {
           zero8(rb[q]);
         }
       }
+      bcol += CBK;
       bcur_k += CBK;
       while (bcur_k >= g.K) {
         bcur_k -= g.K;
-        bcur_rsi += 1;
+        if (++bsct == Sp) {
+          bsct = 0;
+          bcol += b_rw;
+        } else {
+          bcol += b_sw;
+        }
       }
     }
   };
