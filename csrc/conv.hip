@@ -692,8 +692,11 @@ __global__ __launch_bounds__(256) void conv_wgrad_fast_kernel(
     const bf16* __restrict__ x, const bf16* __restrict__ dy,
     float* __restrict__ dw, Geom g, long NP, long npslice) {
   // images: [inner16-block][np(64)][16] per operand (4 blocks of 64 k/rsc)
-  __shared__ __bf16 sA[4 * WIMG];
-  __shared__ __bf16 sB[4 * WIMG];
+  // DOUBLE-buffered: one barrier per np-tile, and the next tile's global
+  // loads issue before this tile's MFMA phase (the single-buffered form
+  // ran 80% WAIT: load latency sat inside the tile, unoverlapped)
+  __shared__ __bf16 sA[2 * 4 * WIMG];
+  __shared__ __bf16 sB[2 * 4 * WIMG];
 
   const long wgid = xcd_tile_remap3();  // z-major chunks per XCD
   const long bx = wgid % gridDim.x;
@@ -725,8 +728,12 @@ __global__ __launch_bounds__(256) void conv_wgrad_fast_kernel(
   }
   const bool a_ok = k0c + scol + 8 <= g.K;
 
-  // incremental pixel cursors for rows snp and snp+32
-  int pn[2], pho[2], pwo[2];
+  // incremental pixel cursors + INCREMENTAL addresses for rows snp, snp+32
+  int pn[2], pho[2], pwo[2], fh[2], fw[2];
+  long aoff[2], xoff[2];
+  const int sC = g.stride * g.C;
+  const long sWC = (long)g.stride * g.W * g.C;
+  const long HWC = (long)g.H * g.W * g.C;
 #pragma unroll
   for (int q = 0; q < 2; ++q) {
     long m = np0 + snp + 32 * q;
@@ -735,6 +742,10 @@ __global__ __launch_bounds__(256) void conv_wgrad_fast_kernel(
     int rem = (int)(mm % ((long)g.Ho * g.Wo));
     pho[q] = rem / g.Wo;
     pwo[q] = rem % g.Wo;
+    fh[q] = pho[q] * g.stride - g.pad + fb_r;
+    fw[q] = pwo[q] * g.stride - g.pad + fb_s;
+    aoff[q] = mm * g.K + k0c + scol;
+    xoff[q] = (((long)pn[q] * g.H + fh[q]) * g.W + fw[q]) * g.C + fb_c;
   }
   // O(1) cursor advance: np += WBK decomposes into per-shape constants
   // (the while-wrap form costs up to ~16 serial VALU iterations per step
@@ -743,77 +754,100 @@ __global__ __launch_bounds__(256) void conv_wgrad_fast_kernel(
   const int dcar = WBK / g.Wo;
   const int dho = dcar % g.Ho;
   const int dn = dcar / g.Ho;
+  const long dA = (long)WBK * g.K;
 
-  f32x4 acc[2][2] = {};
+  __bf16 ra[2][8], rb[2][8];
 
-  for (long p0 = np0; p0 < np1; p0 += WBK) {
-    __bf16 ra[2][8], rb[2][8];
+  auto stage = [&](long p0) {
 #pragma unroll
     for (int q = 0; q < 2; ++q) {
       const long m = p0 + snp + 32 * q;
       const bool v = m < np1;
       if (v && a_ok) {
-        *reinterpret_cast<s16x8*>(ra[q]) = *reinterpret_cast<const s16x8*>(
-            dy + m * g.K + k0c + scol);
+        *reinterpret_cast<s16x8*>(ra[q]) =
+            *reinterpret_cast<const s16x8*>(dy + aoff[q]);
       } else {
         zero8(ra[q]);
       }
-      if (v && bcol_ok) {
-        int h = pho[q] * g.stride - g.pad + fb_r;
-        int w = pwo[q] * g.stride - g.pad + fb_s;
-        if ((unsigned)h < (unsigned)g.H && (unsigned)w < (unsigned)g.W) {
-          long off = (((long)pn[q] * g.H + h) * g.W + w) * g.C + fb_c;
-          *reinterpret_cast<s16x8*>(rb[q]) =
-              *reinterpret_cast<const s16x8*>(x + off);
-        } else {
-          zero8(rb[q]);
-        }
+      if (v && bcol_ok && (unsigned)fh[q] < (unsigned)g.H &&
+          (unsigned)fw[q] < (unsigned)g.W) {
+        *reinterpret_cast<s16x8*>(rb[q]) =
+            *reinterpret_cast<const s16x8*>(x + xoff[q]);
       } else {
         zero8(rb[q]);
       }
-      if (v) {
-        // constant-delta advance (see dwo/dho/dn above): <= 2 cond. wraps
-        pwo[q] += dwo;
-        int car = 0;
-        if (pwo[q] >= g.Wo) {
-          pwo[q] -= g.Wo;
-          car = 1;
-        }
-        pho[q] += dho + car;
-        int car2 = 0;
-        if (pho[q] >= g.Ho) {
-          pho[q] -= g.Ho;
-          car2 = 1;
-        }
-        pn[q] += dn + car2;
+      // constant-delta advance (dwo/dho/dn): <= 2 conditional wraps; the
+      // addresses track the cursors with adds only
+      aoff[q] += dA;
+      pwo[q] += dwo;
+      fw[q] += dwo * g.stride;
+      xoff[q] += (long)dwo * sC;
+      int car = 0;
+      if (pwo[q] >= g.Wo) {
+        pwo[q] -= g.Wo;
+        fw[q] -= g.Wo * g.stride;
+        xoff[q] -= (long)g.Wo * sC;
+        car = 1;
       }
+      pho[q] += dho + car;
+      fh[q] += (dho + car) * g.stride;
+      xoff[q] += (dho + car) * sWC;
+      int car2 = 0;
+      if (pho[q] >= g.Ho) {
+        pho[q] -= g.Ho;
+        fh[q] -= g.Ho * g.stride;
+        xoff[q] -= (long)g.Ho * sWC;
+        car2 = 1;
+      }
+      xoff[q] += (dn + car2) * HWC;
+      pn[q] += dn + car2;
     }
-    __syncthreads();
+  };
+
+  auto write_lds = [&](int pbuf) {
+    __bf16* wA = sA + pbuf * 4 * WIMG;
+    __bf16* wB = sB + pbuf * 4 * WIMG;
 #pragma unroll
     for (int q = 0; q < 2; ++q) {
       const int npl = snp + 32 * q;
-      *reinterpret_cast<bf16x8*>(&sA[simg + np_img(npl)]) =
+      *reinterpret_cast<bf16x8*>(&wA[simg + np_img(npl)]) =
           *reinterpret_cast<bf16x8*>(ra[q]);
-      *reinterpret_cast<bf16x8*>(&sB[simg + np_img(npl)]) =
+      *reinterpret_cast<bf16x8*>(&wB[simg + np_img(npl)]) =
           *reinterpret_cast<bf16x8*>(rb[q]);
     }
+  };
+
+  f32x4 acc[2][2] = {};
+  stage(np0);
+  write_lds(0);
+  if (np0 + WBK < np1) stage(np0 + WBK);
+  int pb = 0;
+
+  for (long p0 = np0; p0 < np1; p0 += WBK) {
     __syncthreads();
+    if (p0 + WBK < np1) {
+      write_lds(pb ^ 1);
+      if (p0 + 2 * WBK < np1) stage(p0 + 2 * WBK);
+    }
+    const __bf16* rA = sA + pb * 4 * WIMG;
+    const __bf16* rB = sB + pb * 4 * WIMG;
 
 #pragma unroll
     for (int ks = 0; ks < 2; ++ks) {
 #pragma unroll
       for (int fi = 0; fi < 2; ++fi) {
         bf16x8 af =
-            tr16_frag(&sA[(wr * 2 + fi) * WIMG], ks * 32, lane);
+            tr16_frag(&rA[(wr * 2 + fi) * WIMG], ks * 32, lane);
 #pragma unroll
         for (int fj = 0; fj < 2; ++fj) {
           bf16x8 bfr =
-              tr16_frag(&sB[(wc * 2 + fj) * WIMG], ks * 32, lane);
+              tr16_frag(&rB[(wc * 2 + fj) * WIMG], ks * 32, lane);
           acc[fi][fj] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               af, bfr, acc[fi][fj], 0, 0, 0);
         }
       }
     }
+    pb ^= 1;
   }
 
   const int cn = lane & 15;
