@@ -101,3 +101,83 @@ def test_decide_matches_step_mask():
             d = ctrl.decide(norms, p)
             f = ctrl.step(norms, p)
             assert (d == f).all(), (adaptive, p)
+
+
+def test_engine_mask_lookahead_protocol():
+    """engine.step() must post the NEXT pass's mask exchange (decide-only,
+    no state commit) and begin_pass must commit + wait — with the commit
+    mask identical to the posted one, and no posts during warmup/decent
+    (static all-fire)."""
+    import numpy as np
+    import torch
+    from eventgrad_amd.config import RunConfig, TriggerConfig, OptimConfig
+    from eventgrad_amd.models import build_model
+    from eventgrad_amd.parallel.engine import GossipEngine
+    from eventgrad_amd.parallel.flat import FlatParamSpace
+
+    class RecordingTransport:
+        def __init__(self):
+            self.log = []
+            self._mask = None
+
+        def post_masks(self, mask):
+            self.log.append(("post", mask.clone()))
+            self._mask = mask.cpu()
+
+        def wait_masks(self):
+            self.log.append(("wait",))
+            m, self._mask = self._mask, None
+            return m.clone(), m.clone()
+
+        def exchange_masks(self, mask):
+            self.log.append(("blocking", mask.clone()))
+            m = mask.cpu()
+            return m.clone(), m.clone()
+
+        def post_payloads(self, *a):
+            self.log.append(("payloads",))
+
+        def finish(self):
+            pass
+
+        def cancel_pending_masks(self):
+            self.log.append(("cancel",))
+
+    torch.manual_seed(0)
+    model = build_model("mlp")
+    dev = torch.device("cpu")
+    space = FlatParamSpace(model, dev)
+    cfg = RunConfig(mode="event", device="cpu",
+                    optim=OptimConfig(lr=0.05),
+                    trigger=TriggerConfig(adaptive=True, horizon=1.01,
+                                          initial_comm_passes=3))
+    eng = GossipEngine(space, cfg, rank=0, world=2, device=dev)
+    tr = RecordingTransport()
+    eng.transport = tr
+
+    x = torch.randn(8, 1, 28, 28)
+    y = torch.randint(0, 10, (8,))
+    from eventgrad_amd.ops import functional as O
+    for p in range(1, 8):
+        eng.begin_pass(p)
+        space.zero_grad()
+        loss = O.nll_of_logits(model(x), y)
+        loss.backward()
+        eng.after_backward()
+        eng.step()
+
+    ops = [e[0] for e in tr.log]
+    # passes 1-2 are warmup (static all-fire): no mask traffic at all
+    first_post = ops.index("post")
+    assert "blocking" not in ops[:first_post]
+    # lookahead posted at step(p) for p+1 >= warmup; begin_pass waits.
+    # The LAST step's post is for a pass that never runs — finalize drains
+    # it (cancel path), so posts == waits + 1 with a pending decision left.
+    assert eng._pending is not None
+    assert ops.count("post") == ops.count("wait") + 1
+    # strict alternation post -> wait (never two posts in flight)
+    seq = [o for o in ops if o in ("post", "wait")]
+    assert all(a == "post" and b == "wait"
+               for a, b in zip(seq[::2], seq[1::2]))
+    eng.transport.cancel_pending_masks()
+    assert tr.log[-1][0] == "cancel"
