@@ -110,6 +110,8 @@ def main():
     ap.add_argument("--outdir", default="benchmarks/out")
     ap.add_argument("--launched", action="store_true",
                     help="already under torchrun; do not self-spawn")
+    ap.add_argument("--port", type=int, default=29650,
+                    help="rendezvous base port (change for concurrent runs)")
     args = ap.parse_args()
     os.makedirs(args.outdir, exist_ok=True)
 
@@ -118,7 +120,7 @@ def main():
             run_one(mode, args, os.path.join(args.outdir, f"{mode}.json"))
     else:
         mp.start_processes(worker,
-                           args=(args.world, 29650, vars(args), args.outdir),
+                           args=(args.world, args.port, vars(args), args.outdir),
                            nprocs=args.world, start_method="spawn", join=True)
 
     results = {}
