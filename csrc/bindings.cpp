@@ -51,7 +51,10 @@ std::vector<torch::Tensor> bn_fwd(torch::Tensor, torch::Tensor, torch::Tensor,
 std::vector<torch::Tensor> bn_bwd(torch::Tensor, torch::Tensor, torch::Tensor,
                                   torch::Tensor, torch::Tensor, torch::Tensor,
                                   bool, bool, c10::optional<torch::Tensor>,
-                                  c10::optional<torch::Tensor>);
+                                  c10::optional<torch::Tensor>, bool);
+torch::Tensor relu_bwd_bnstats(torch::Tensor, torch::Tensor, torch::Tensor,
+                               torch::Tensor, torch::Tensor, torch::Tensor,
+                               torch::Tensor);
 // pool.hip
 std::vector<torch::Tensor> maxpool2x2_fwd(torch::Tensor);
 torch::Tensor maxpool2x2_bwd(torch::Tensor, torch::Tensor, long, long);
@@ -105,7 +108,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("invstd"), py::arg("gamma"), py::arg("y"), py::arg("relu"),
         py::arg("training"),
         py::arg("dgamma_out") = c10::nullopt,
-        py::arg("dbeta_out") = c10::nullopt);
+        py::arg("dbeta_out") = c10::nullopt,
+        py::arg("stats_ready") = false);
+  m.def("relu_bwd_bnstats", &eg::relu_bwd_bnstats);
   m.def("maxpool2x2_fwd", &eg::maxpool2x2_fwd);
   m.def("maxpool2x2_bwd", &eg::maxpool2x2_bwd);
   m.def("avgpool_fwd", &eg::avgpool_fwd);

@@ -241,6 +241,64 @@ __global__ void bwd_stats_kernel(const bf16* __restrict__ dy,
   atomicAdd(&sum_dyx[ch], b);
 }
 
+// fused add_relu backward + the FOLLOWING-in-backward BN's stats: writes
+// da = dy * (y > 0) AND accumulates sum(da), sum(da * xhat) for the BN
+// that PRODUCED the add_relu's first operand (its backward then skips the
+// standalone bwd_stats pass over the same arrays). xb/mean/invstd are that
+// BN's saved input/batch stats; the sums land in its pre-zeroed
+// dgamma/dbeta flat-grad views (direct-grad convention).
+__global__ void relu_bwd_bnstats_kernel(
+    const bf16* __restrict__ dy, const bf16* __restrict__ y,
+    const bf16* __restrict__ xb, const float* __restrict__ mean,
+    const float* __restrict__ invstd, bf16* __restrict__ da,
+    float* __restrict__ sum_dy, float* __restrict__ sum_dyx, long rows,
+    int c) {
+  __shared__ float ls[BN_MAXC], lsq[BN_MAXC];
+  for (int i = threadIdx.x; i < c; i += blockDim.x) {
+    ls[i] = 0.f;
+    lsq[i] = 0.f;
+  }
+  __syncthreads();
+  const int lpr = c >> 3;
+  const int c0 = (threadIdx.x % lpr) * 8;
+  const int rpb = blockDim.x / lpr;
+  float m[8], is[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    m[j] = mean[c0 + j];
+    is[j] = invstd[c0 + j];
+  }
+  float a[8] = {}, b[8] = {};
+  for (long r = (long)blockIdx.x * rpb + threadIdx.x / lpr; r < rows;
+       r += (long)gridDim.x * rpb) {
+    long base = r * c + c0;
+    s16x8 vg = *reinterpret_cast<const s16x8*>(&dy[base]);
+    s16x8 vy = *reinterpret_cast<const s16x8*>(&y[base]);
+    s16x8 vx = *reinterpret_cast<const s16x8*>(&xb[base]);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float g = b2f(__ushort_as_bfloat16((unsigned short)vg[j]));
+      if (b2f(__ushort_as_bfloat16((unsigned short)vy[j])) <= 0.f) g = 0.f;
+      vg[j] = (short)__bfloat16_as_ushort(f2b(g));
+      float xh =
+          (b2f(__ushort_as_bfloat16((unsigned short)vx[j])) - m[j]) * is[j];
+      a[j] += g;
+      b[j] += g * xh;
+    }
+    *reinterpret_cast<s16x8*>(&da[base]) = vg;
+  }
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    atomicAdd(&ls[c0 + j], a[j]);
+    atomicAdd(&lsq[c0 + j], b[j]);
+  }
+  __syncthreads();
+  for (int i = threadIdx.x; i < c; i += blockDim.x) {
+    atomicAdd(&sum_dy[i], ls[i]);
+    atomicAdd(&sum_dyx[i], lsq[i]);
+  }
+}
+
 __global__ void bwd_dx_kernel_v(const bf16* __restrict__ dy,
                                 const bf16* __restrict__ x,
                                 const bf16* __restrict__ y,
@@ -391,12 +449,35 @@ std::vector<torch::Tensor> bn_fwd(torch::Tensor x, torch::Tensor gamma,
   return {y, mean, invstd};
 }
 
+// da = dy * (y > 0), plus the producing BN's backward stats accumulated
+// into its dgamma/dbeta grad views (see relu_bwd_bnstats_kernel)
+torch::Tensor relu_bwd_bnstats(torch::Tensor dy, torch::Tensor y,
+                               torch::Tensor xb, torch::Tensor mean,
+                               torch::Tensor invstd, torch::Tensor dgamma,
+                               torch::Tensor dbeta) {
+  CHECK_IN(dy); CHECK_IN(y); CHECK_IN(xb);
+  CHECK_IN(dgamma); CHECK_IN(dbeta);
+  int c = (int)xb.size(-1);
+  long rows = xb.numel() / c;
+  TORCH_CHECK(c % 8 == 0 && c <= bn::BN_MAXC);
+  auto da = torch::empty_like(dy);
+  int rpb = 256 / (c / 8);
+  int grid = (int)std::min<long>((rows + rpb - 1) / rpb, 256L);
+  bn::relu_bwd_bnstats_kernel<<<grid, 256, 0, cur_stream()>>>(
+      (const bf16*)dy.data_ptr(), (const bf16*)y.data_ptr(),
+      (const bf16*)xb.data_ptr(), mean.data_ptr<float>(),
+      invstd.data_ptr<float>(), (bf16*)da.data_ptr(),
+      dbeta.data_ptr<float>(), dgamma.data_ptr<float>(), rows, c);
+  return da;
+}
+
 std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor x,
                                   torch::Tensor mean, torch::Tensor invstd,
                                   torch::Tensor gamma, torch::Tensor y,
                                   bool relu, bool training,
                                   c10::optional<torch::Tensor> dgamma_out,
-                                  c10::optional<torch::Tensor> dbeta_out) {
+                                  c10::optional<torch::Tensor> dbeta_out,
+                                  bool stats_ready) {
   CHECK_IN(dy); CHECK_IN(x);
   int c = (int)x.size(-1);
   long rows = x.numel() / c;
@@ -418,7 +499,11 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor x,
     sum_dyx = sums.narrow(0, c, c);
   }
   auto stream = cur_stream();
-  if (c % 8 == 0 && c <= bn::BN_MAXC) {
+  if (stats_ready) {
+    // sums already accumulated into the supplied grad views by the
+    // consumer's fused backward kernel (relu_bwd_bnstats)
+    TORCH_CHECK(direct, "stats_ready requires dgamma/dbeta views");
+  } else if (c % 8 == 0 && c <= bn::BN_MAXC) {
     int rpb = 256 / (c / 8);
     int grid = (int)std::min<long>((rows + rpb - 1) / rpb, 256L);
     bn::bwd_stats_kernel_v<<<grid, 256, 0, stream>>>(

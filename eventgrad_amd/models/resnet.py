@@ -56,9 +56,11 @@ class BasicBlock(nn.Module):
     def forward(self, x):
         out = self.bn1(self.conv1(x, bn_stats=True), fuse_relu=True,
                        stats_ready=True)
-        out = self.bn2(self.conv2(out, bn_stats=True), stats_ready=True)
+        out = self.conv2(out, bn_stats=True)
         residual = self.downsampler(x) if self.downsampler is not None else x
-        return O.add_relu(out, residual)
+        # block tail: bn2 -> add+relu with bn2's backward stats fused into
+        # the add_relu backward kernel (ops/functional.py::bn_add_relu)
+        return O.bn_add_relu(self.bn2, out, residual, stats_ready=True)
 
 
 class BottleNeck(nn.Module):
@@ -80,9 +82,9 @@ class BottleNeck(nn.Module):
                        stats_ready=True)
         out = self.bn2(self.conv2(out, bn_stats=True), fuse_relu=True,
                        stats_ready=True)
-        out = self.bn3(self.conv3(out, bn_stats=True), stats_ready=True)
+        out = self.conv3(out, bn_stats=True)
         residual = self.downsampler(x) if self.downsampler is not None else x
-        return O.add_relu(out, residual)
+        return O.bn_add_relu(self.bn3, out, residual, stats_ready=True)
 
 
 class CifarResNet(nn.Module):

@@ -511,3 +511,32 @@ def test_sgd_step_norm_weight_decay():
     # pad gaps get wd*0 contributions only where param is 0 -> identical
     assert rel_err(flat.cpu() * mask.cpu(), p_ref * mask.cpu()) < 1e-6
     assert rel_err(mom.cpu() * mask.cpu(), m_ref * mask.cpu()) < 1e-6
+
+
+def test_relu_bwd_bnstats_fused():
+    """Fused add_relu backward + producing-BN stats == the separate
+    relu_bwd + sum(da), sum(da*xhat) computed in fp32 torch."""
+    torch.manual_seed(4)
+    rows, c = 1024, 64
+    dy = torch.randn(rows, c, device=DEV).to(torch.bfloat16)
+    y = torch.randn(rows, c, device=DEV).to(torch.bfloat16)   # relu out sign
+    xb = torch.randn(rows, c, device=DEV).to(torch.bfloat16)
+    mean = torch.randn(c, device=DEV)
+    invstd = torch.rand(c, device=DEV) + 0.5
+    dgamma = torch.zeros(c, device=DEV)
+    dbeta = torch.zeros(c, device=DEV)
+    da = core().relu_bwd_bnstats(dy, y, xb, mean, invstd, dgamma, dbeta)
+    da_ref = torch.where(y.float() > 0, dy.float(), torch.zeros(1, device=DEV))
+    assert torch.equal(da.float(), da_ref.to(torch.bfloat16).float())
+    xh = (xb.float() - mean) * invstd
+    np.testing.assert_allclose(dbeta.cpu().numpy(),
+                               da_ref.sum(0).cpu().numpy(), rtol=1e-4,
+                               atol=1e-3)
+    np.testing.assert_allclose(dgamma.cpu().numpy(),
+                               (da_ref * xh).sum(0).cpu().numpy(), rtol=1e-3,
+                               atol=1e-2)
+    # accumulate semantics: second call adds
+    core().relu_bwd_bnstats(dy, y, xb, mean, invstd, dgamma, dbeta)
+    np.testing.assert_allclose(dbeta.cpu().numpy(),
+                               2 * da_ref.sum(0).cpu().numpy(), rtol=1e-4,
+                               atol=2e-3)
