@@ -168,7 +168,7 @@ def conv2d(x, w, bias=None, stride=1, padding=0, bn_stats=False,
 class _BatchNormNHWC(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, gamma, beta, running_mean, running_var, training,
-                momentum, eps, fuse_relu, stats_ready):
+                momentum, eps, fuse_relu, stats_ready, stash_module):
         core = native()
         y, save_mean, save_invstd = core.bn_fwd(
             x, gamma.detach(), beta.detach(), running_mean, running_var,
@@ -181,6 +181,12 @@ class _BatchNormNHWC(torch.autograd.Function):
         ctx.bgrad = beta.grad if beta.requires_grad else None
         ctx.fuse_relu = fuse_relu
         ctx.training = training
+        # stats_external: the CONSUMER's fused backward (bn_add_relu ->
+        # relu_bwd_bnstats) accumulates sum(da), sum(da*xhat) into the
+        # grad views before this backward runs; skip the stats pass here
+        ctx.stats_external = stash_module is not None
+        if stash_module is not None:
+            stash_module._bwd_stash = (x, save_mean, save_invstd)
         return y
 
     @staticmethod
@@ -194,22 +200,24 @@ class _BatchNormNHWC(torch.autograd.Function):
                                    save_invstd, gamma.detach(), y,
                                    ctx.fuse_relu, ctx.training,
                                    dgamma_out=ctx.ggrad,
-                                   dbeta_out=ctx.bgrad)
+                                   dbeta_out=ctx.bgrad,
+                                   stats_ready=ctx.stats_external)
             return (dx, None, None, None, None, None, None, None, None,
-                    None)
+                    None, None)
         dx, dgamma, dbeta = core.bn_bwd(dy.contiguous(), x, save_mean,
                                         save_invstd, gamma.detach(), y,
                                         ctx.fuse_relu, ctx.training)
         return (dx, dgamma, dbeta, None, None, None, None, None, None,
-                None)
+                None, None)
 
 
 def batch_norm(x, gamma, beta, running_mean, running_var, training,
-               momentum=0.1, eps=1e-5, fuse_relu=False, stats_ready=False):
+               momentum=0.1, eps=1e-5, fuse_relu=False, stats_ready=False,
+               stash_module=None):
     if use_native(x):
         return _BatchNormNHWC.apply(x, gamma, beta, running_mean, running_var,
                                     training, momentum, eps, fuse_relu,
-                                    stats_ready and training)
+                                    stats_ready and training, stash_module)
     y = F.batch_norm(x, running_mean, running_var, gamma, beta, training,
                      momentum, eps)
     return F.relu(y) if fuse_relu else y
@@ -238,23 +246,53 @@ def relu(x):
 
 class _AddReLU(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, a, b):
+    def forward(ctx, a, b, xb, mean, invstd, ggrad, bgrad):
         y = native().add_relu_fwd(a, b)
-        ctx.save_for_backward(y)
+        if xb is not None:
+            ctx.save_for_backward(y, xb, mean, invstd)
+            ctx.ggrad, ctx.bgrad = ggrad, bgrad
+        else:
+            ctx.save_for_backward(y)
+            ctx.ggrad = None
         return y
 
     @staticmethod
     def backward(ctx, dy):
-        (y,) = ctx.saved_tensors
-        da = native().relu_bwd(dy.contiguous(), y)
-        return da, da
+        if ctx.ggrad is not None:
+            # fused: da = dy * (y > 0) AND the producing BN's backward
+            # stats accumulate into its grad views in the same pass
+            y, xb, mean, invstd = ctx.saved_tensors
+            da = native().relu_bwd_bnstats(dy.contiguous(), y, xb, mean,
+                                           invstd, ctx.ggrad, ctx.bgrad)
+        else:
+            (y,) = ctx.saved_tensors
+            da = native().relu_bwd(dy.contiguous(), y)
+        return da, da, None, None, None, None, None
 
 
 def add_relu(a, b):
     """Residual join: relu(a + b) (resnet.hpp:46-48)."""
     if use_native(a):
-        return _AddReLU.apply(a, b)
+        return _AddReLU.apply(a, b, None, None, None, None, None)
     return F.relu(a + b)
+
+
+def bn_add_relu(bn, x, residual, stats_ready=False):
+    """BatchNorm (no inline relu) -> residual add+relu, the block-tail
+    pattern (resnet.hpp:41-48), with the BN's BACKWARD stats fused into
+    the add_relu backward kernel when eligible: the standalone
+    bwd_stats pass over (da, x) disappears. `bn` is the BatchNorm2d
+    module; eligibility = native path + training + flat-space grad views
+    + C % 8 == 0, C <= 1024 (kernel constraints)."""
+    fuse = (use_native(x) and bn.training
+            and bn.weight.grad is not None and bn.bias.grad is not None
+            and bn.num_features % 8 == 0 and bn.num_features <= 1024)
+    out = bn(x, stats_ready=stats_ready, stats_external=fuse)
+    if fuse:
+        xb, mean, invstd = bn._bwd_stash
+        return _AddReLU.apply(out, residual, xb, mean, invstd,
+                              bn.weight.grad, bn.bias.grad)
+    return add_relu(out, residual)
 
 
 class _Dropout(torch.autograd.Function):
