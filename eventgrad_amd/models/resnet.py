@@ -56,8 +56,11 @@ class BasicBlock(nn.Module):
     def forward(self, x):
         out = self.bn1(self.conv1(x, bn_stats=True), fuse_relu=True,
                        stats_ready=True)
-        out = self.conv2(out, bn_stats=True)
+        # residual FIRST: the downsampler's conv+BN use the same shared
+        # per-channel-count stats workspace as the conv2->bn2 pair below —
+        # interleaving them corrupts both stats streams
         residual = self.downsampler(x) if self.downsampler is not None else x
+        out = self.conv2(out, bn_stats=True)
         # block tail: bn2 -> add+relu with bn2's backward stats fused into
         # the add_relu backward kernel (ops/functional.py::bn_add_relu)
         return O.bn_add_relu(self.bn2, out, residual, stats_ready=True)
@@ -82,8 +85,8 @@ class BottleNeck(nn.Module):
                        stats_ready=True)
         out = self.bn2(self.conv2(out, bn_stats=True), fuse_relu=True,
                        stats_ready=True)
-        out = self.conv3(out, bn_stats=True)
         residual = self.downsampler(x) if self.downsampler is not None else x
+        out = self.conv3(out, bn_stats=True)
         return O.bn_add_relu(self.bn3, out, residual, stats_ready=True)
 
 
