@@ -461,9 +461,12 @@ torch::Tensor relu_bwd_bnstats(torch::Tensor dy, torch::Tensor y,
   long rows = xb.numel() / c;
   TORCH_CHECK(c % 8 == 0 && c <= bn::BN_MAXC);
   auto da = torch::empty_like(dy);
-  int rpb = 256 / (c / 8);
+  // 512-thread blocks: this kernel also carries the elementwise da write,
+  // so it wants more in-flight rows per block than the pure reductions
+  // (1024 blocks x 256 thr measured slower: 4x the global atomic merges)
+  int rpb = 512 / (c / 8);
   int grid = (int)std::min<long>((rows + rpb - 1) / rpb, 256L);
-  bn::relu_bwd_bnstats_kernel<<<grid, 256, 0, cur_stream()>>>(
+  bn::relu_bwd_bnstats_kernel<<<grid, 512, 0, cur_stream()>>>(
       (const bf16*)dy.data_ptr(), (const bf16*)y.data_ptr(),
       (const bf16*)xb.data_ptr(), mean.data_ptr<float>(),
       invstd.data_ptr<float>(), (bf16*)da.data_ptr(),
