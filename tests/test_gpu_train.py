@@ -352,3 +352,36 @@ def test_direct_grads_match_fallback(model_name, mode):
         # order differs
         err = (gr - gf).norm().item() / denom
         assert err < 1e-4, (name, err)
+
+
+def test_bn_tail_fusion_grads_exact():
+    """The bn_add_relu fused backward (relu_bwd_bnstats) must deposit
+    exactly sum(da) / sum(da*xhat) into the BN's grad views, verified
+    against fp32 math recomputed from the SAME run's saved stats."""
+    from eventgrad_amd.models.resnet import BasicBlock
+    from eventgrad_amd.parallel.flat import FlatParamSpace
+
+    dev = torch.device("cuda")
+    torch.manual_seed(9)
+    block = BasicBlock(16, 16).to(dev)
+    block.train()
+    space = FlatParamSpace(block, dev)  # grad views -> fusion eligible
+
+    x = torch.randn(8, 16, 16, 16, device=dev) \
+        .permute(0, 2, 3, 1).contiguous().to(torch.bfloat16)
+    space.zero_grad()
+    y = block(x)
+    y.float().sum().backward()      # dy of the block output == 1
+    torch.cuda.synchronize()
+
+    assert hasattr(block.bn2, "_bwd_stash"), "fusion did not engage"
+    xb, mean, invstd = block.bn2._bwd_stash
+    da = (y.detach().float() > 0).float()        # dy=1 gated by relu out
+    xh = (xb.float() - mean) * invstd
+    C = xb.shape[-1]
+    dbeta_ref = da.reshape(-1, C).sum(0)
+    dgamma_ref = (da * xh).reshape(-1, C).sum(0)
+    np.testing.assert_allclose(block.bn2.bias.grad.cpu().numpy(),
+                               dbeta_ref.cpu().numpy(), rtol=1e-4, atol=1e-2)
+    np.testing.assert_allclose(block.bn2.weight.grad.cpu().numpy(),
+                               dgamma_ref.cpu().numpy(), rtol=1e-3, atol=5e-2)
