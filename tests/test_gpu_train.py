@@ -377,7 +377,7 @@ def test_bn_tail_fusion_grads_exact():
     assert hasattr(block.bn2, "_bwd_stash"), "fusion did not engage"
     xb, mean, invstd = block.bn2._bwd_stash
     da = (y.detach().float() > 0).float()        # dy=1 gated by relu out
-    xh = (xb.float() - mean) * invstd
+    xh = (xb.detach().float() - mean) * invstd
     C = xb.shape[-1]
     dbeta_ref = da.reshape(-1, C).sum(0)
     dgamma_ref = (da * xh).reshape(-1, C).sum(0)
