@@ -55,6 +55,11 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor, torch::Tensor, torch::Tensor,
 torch::Tensor relu_bwd_bnstats(torch::Tensor, torch::Tensor, torch::Tensor,
                                torch::Tensor, torch::Tensor, torch::Tensor,
                                torch::Tensor);
+std::vector<torch::Tensor> bn_stats_finalize(torch::Tensor, torch::Tensor,
+                                             torch::Tensor, double, double,
+                                             bool);
+torch::Tensor bn_norm_add_relu(torch::Tensor, torch::Tensor, torch::Tensor,
+                               torch::Tensor, torch::Tensor, torch::Tensor);
 // pool.hip
 std::vector<torch::Tensor> maxpool2x2_fwd(torch::Tensor);
 torch::Tensor maxpool2x2_bwd(torch::Tensor, torch::Tensor, long, long);
@@ -111,6 +116,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("dbeta_out") = c10::nullopt,
         py::arg("stats_ready") = false);
   m.def("relu_bwd_bnstats", &eg::relu_bwd_bnstats);
+  m.def("bn_stats_finalize", &eg::bn_stats_finalize);
+  m.def("bn_norm_add_relu", &eg::bn_norm_add_relu);
   m.def("maxpool2x2_fwd", &eg::maxpool2x2_fwd);
   m.def("maxpool2x2_bwd", &eg::maxpool2x2_bwd);
   m.def("avgpool_fwd", &eg::avgpool_fwd);

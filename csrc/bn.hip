@@ -159,6 +159,34 @@ __global__ void norm_kernel(const bf16* __restrict__ x, bf16* __restrict__ y,
   }
 }
 
+// fused block-tail forward: out = relu(xhat*gamma + beta + residual) —
+// the separate normalize kernel's y2 write and the add_relu's y2 read
+// disappear (y2 is never needed in backward: the BN bwd gating uses the
+// RELU output and its dx path uses x2)
+__global__ void norm_add_relu_kernel_v(
+    const bf16* __restrict__ x, const bf16* __restrict__ res,
+    bf16* __restrict__ out, const float* __restrict__ mean,
+    const float* __restrict__ invstd, const float* __restrict__ gamma,
+    const float* __restrict__ beta, long n8, int c8) {
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (long v = (long)blockIdx.x * blockDim.x + threadIdx.x; v < n8;
+       v += stride) {
+    int c0 = (int)(v % c8) * 8;
+    s16x8 xv = reinterpret_cast<const s16x8*>(x)[v];
+    s16x8 rv = reinterpret_cast<const s16x8*>(res)[v];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      int ch = c0 + j;
+      float f = (b2f(__ushort_as_bfloat16((unsigned short)xv[j])) - mean[ch])
+                * invstd[ch] * gamma[ch] + beta[ch]
+                + b2f(__ushort_as_bfloat16((unsigned short)rv[j]));
+      if (f < 0.f) f = 0.f;
+      xv[j] = (short)__bfloat16_as_ushort(f2b(f));
+    }
+    reinterpret_cast<s16x8*>(out)[v] = xv;
+  }
+}
+
 // vectorized backward reductions (C % 8 == 0): sum(dy'), sum(dy'*xhat)
 __global__ void bwd_stats_kernel_v(const bf16* __restrict__ dy,
                                    const bf16* __restrict__ x,
@@ -447,6 +475,63 @@ std::vector<torch::Tensor> bn_fwd(torch::Tensor x, torch::Tensor gamma,
         gamma.data_ptr<float>(), beta.data_ptr<float>(), n, c, relu ? 1 : 0);
   }
   return {y, mean, invstd};
+}
+
+// batch stats + running-stats update WITHOUT the normalize pass: returns
+// (mean, invstd). have_stats: the producing conv's epilogue already
+// filled the shared workspace.
+std::vector<torch::Tensor> bn_stats_finalize(torch::Tensor x,
+                                             torch::Tensor running_mean,
+                                             torch::Tensor running_var,
+                                             double momentum, double eps,
+                                             bool have_stats) {
+  CHECK_IN(x);
+  int c = (int)x.size(-1);
+  long rows = x.numel() / c;
+  auto f32 = x.options().dtype(torch::kFloat32);
+  auto mean = torch::empty({c}, f32);
+  auto invstd = torch::empty({c}, f32);
+  auto stream = cur_stream();
+  auto ssq = bn_ws_tensor(c, f32);
+  auto sum = ssq.narrow(0, 0, c);
+  auto sq = ssq.narrow(0, c, c);
+  if (have_stats) {
+    // already accumulated by the conv epilogue
+  } else if (c % 8 == 0 && c <= bn::BN_MAXC) {
+    int rpb = 256 / (c / 8);
+    int grid = (int)std::min<long>((rows + rpb - 1) / rpb, 256L);
+    bn::stats_kernel_v<<<grid, 256, 0, stream>>>(
+        (const bf16*)x.data_ptr(), sum.data_ptr<float>(),
+        sq.data_ptr<float>(), rows, c);
+  } else {
+    dim3 grid((c + 63) / 64, (unsigned)std::min<long>((rows + 3) / 4, 256L));
+    bn::stats_kernel<<<grid, 256, 0, stream>>>(
+        (const bf16*)x.data_ptr(), sum.data_ptr<float>(),
+        sq.data_ptr<float>(), rows, c);
+  }
+  bn::finalize_kernel<<<ceil_div(c, 128), 128, 0, stream>>>(
+      sum.data_ptr<float>(), sq.data_ptr<float>(), mean.data_ptr<float>(),
+      invstd.data_ptr<float>(), running_mean.data_ptr<float>(),
+      running_var.data_ptr<float>(), rows, c, (float)momentum, (float)eps,
+      1);
+  return {mean, invstd};
+}
+
+torch::Tensor bn_norm_add_relu(torch::Tensor x, torch::Tensor res,
+                               torch::Tensor mean, torch::Tensor invstd,
+                               torch::Tensor gamma, torch::Tensor beta) {
+  CHECK_IN(x); CHECK_IN(res);
+  int c = (int)x.size(-1);
+  TORCH_CHECK(c % 8 == 0);
+  auto out = torch::empty_like(x);
+  long n8 = x.numel() / 8;
+  int grid = (int)std::min<long>((n8 + 255) / 256, 4096L);
+  bn::norm_add_relu_kernel_v<<<grid, 256, 0, cur_stream()>>>(
+      (const bf16*)x.data_ptr(), (const bf16*)res.data_ptr(),
+      (bf16*)out.data_ptr(), mean.data_ptr<float>(),
+      invstd.data_ptr<float>(), gamma.data_ptr<float>(),
+      beta.data_ptr<float>(), n8, c / 8);
+  return out;
 }
 
 // da = dy * (y > 0), plus the producing BN's backward stats accumulated

@@ -80,16 +80,14 @@ class BatchNorm2d(nn.Module):
     def sync_buffers_after_load(self):
         self._nbt = int(self.num_batches_tracked.item())
 
-    def forward(self, x, fuse_relu: bool = False, stats_ready: bool = False,
-                stats_external: bool = False):
+    def forward(self, x, fuse_relu: bool = False, stats_ready: bool = False):
         if self.training:
             self._nbt += 1
         return O.batch_norm(x, self.weight, self.bias, self.running_mean,
                             self.running_var, self.training, self.momentum,
                             self.eps, fuse_relu,
                             stats_ready and O.use_native(x) and self.training
-                            and self.num_features % 64 == 0,
-                            stash_module=self if stats_external else None)
+                            and self.num_features % 64 == 0)
 
 
 class Linear(nn.Module):

@@ -168,7 +168,7 @@ def conv2d(x, w, bias=None, stride=1, padding=0, bn_stats=False,
 class _BatchNormNHWC(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, gamma, beta, running_mean, running_var, training,
-                momentum, eps, fuse_relu, stats_ready, stash_module):
+                momentum, eps, fuse_relu, stats_ready):
         core = native()
         y, save_mean, save_invstd = core.bn_fwd(
             x, gamma.detach(), beta.detach(), running_mean, running_var,
@@ -181,12 +181,6 @@ class _BatchNormNHWC(torch.autograd.Function):
         ctx.bgrad = beta.grad if beta.requires_grad else None
         ctx.fuse_relu = fuse_relu
         ctx.training = training
-        # stats_external: the CONSUMER's fused backward (bn_add_relu ->
-        # relu_bwd_bnstats) accumulates sum(da), sum(da*xhat) into the
-        # grad views before this backward runs; skip the stats pass here
-        ctx.stats_external = stash_module is not None
-        if stash_module is not None:
-            stash_module._bwd_stash = (x, save_mean, save_invstd)
         return y
 
     @staticmethod
@@ -200,24 +194,22 @@ class _BatchNormNHWC(torch.autograd.Function):
                                    save_invstd, gamma.detach(), y,
                                    ctx.fuse_relu, ctx.training,
                                    dgamma_out=ctx.ggrad,
-                                   dbeta_out=ctx.bgrad,
-                                   stats_ready=ctx.stats_external)
+                                   dbeta_out=ctx.bgrad)
             return (dx, None, None, None, None, None, None, None, None,
-                    None, None)
+                    None)
         dx, dgamma, dbeta = core.bn_bwd(dy.contiguous(), x, save_mean,
                                         save_invstd, gamma.detach(), y,
                                         ctx.fuse_relu, ctx.training)
         return (dx, dgamma, dbeta, None, None, None, None, None, None,
-                None, None)
+                None)
 
 
 def batch_norm(x, gamma, beta, running_mean, running_var, training,
-               momentum=0.1, eps=1e-5, fuse_relu=False, stats_ready=False,
-               stash_module=None):
+               momentum=0.1, eps=1e-5, fuse_relu=False, stats_ready=False):
     if use_native(x):
         return _BatchNormNHWC.apply(x, gamma, beta, running_mean, running_var,
                                     training, momentum, eps, fuse_relu,
-                                    stats_ready and training, stash_module)
+                                    stats_ready and training)
     y = F.batch_norm(x, running_mean, running_var, gamma, beta, training,
                      momentum, eps)
     return F.relu(y) if fuse_relu else y
@@ -246,52 +238,78 @@ def relu(x):
 
 class _AddReLU(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, a, b, xb, mean, invstd, ggrad, bgrad):
+    def forward(ctx, a, b):
         y = native().add_relu_fwd(a, b)
-        if xb is not None:
-            ctx.save_for_backward(y, xb, mean, invstd)
-            ctx.ggrad, ctx.bgrad = ggrad, bgrad
-        else:
-            ctx.save_for_backward(y)
-            ctx.ggrad = None
+        ctx.save_for_backward(y)
         return y
 
     @staticmethod
     def backward(ctx, dy):
-        if ctx.ggrad is not None:
-            # fused: da = dy * (y > 0) AND the producing BN's backward
-            # stats accumulate into its grad views in the same pass
-            y, xb, mean, invstd = ctx.saved_tensors
-            da = native().relu_bwd_bnstats(dy.contiguous(), y, xb, mean,
-                                           invstd, ctx.ggrad, ctx.bgrad)
-        else:
-            (y,) = ctx.saved_tensors
-            da = native().relu_bwd(dy.contiguous(), y)
-        return da, da, None, None, None, None, None
+        (y,) = ctx.saved_tensors
+        da = native().relu_bwd(dy.contiguous(), y)
+        return da, da
 
 
 def add_relu(a, b):
     """Residual join: relu(a + b) (resnet.hpp:46-48)."""
     if use_native(a):
-        return _AddReLU.apply(a, b, None, None, None, None, None)
+        return _AddReLU.apply(a, b)
     return F.relu(a + b)
+
+
+class _BnTail(torch.autograd.Function):
+    """Whole block tail as one autograd node: batch-stats finalize ->
+    out = relu(xhat*gamma + beta + residual) in ONE elementwise kernel
+    (the normalize pass's y2 write and the add_relu's y2 read disappear;
+    y2 is never needed in backward). Backward: relu_bwd_bnstats (da +
+    dgamma/dbeta into the grad views) then bn dx with stats_ready."""
+
+    @staticmethod
+    def forward(ctx, x, residual, gamma, beta, running_mean, running_var,
+                momentum, eps, have_stats, ggrad, bgrad, stash_module):
+        core = native()
+        mean, invstd = core.bn_stats_finalize(x, running_mean, running_var,
+                                              momentum, eps, have_stats)
+        out = core.bn_norm_add_relu(x, residual, mean, invstd,
+                                    gamma.detach(), beta.detach())
+        ctx.save_for_backward(x, gamma, mean, invstd, out)
+        ctx.ggrad, ctx.bgrad = ggrad, bgrad
+        if stash_module is not None:
+            stash_module._bwd_stash = (x, mean, invstd)
+        return out
+
+    @staticmethod
+    def backward(ctx, dout):
+        core = native()
+        x, gamma, mean, invstd, out = ctx.saved_tensors
+        da = core.relu_bwd_bnstats(dout.contiguous(), out, x, mean, invstd,
+                                   ctx.ggrad, ctx.bgrad)
+        dx, _, _ = core.bn_bwd(da, x, mean, invstd, gamma.detach(), out,
+                               False, True, dgamma_out=ctx.ggrad,
+                               dbeta_out=ctx.bgrad, stats_ready=True)
+        return (dx, da, None, None, None, None, None, None, None, None,
+                None, None)
 
 
 def bn_add_relu(bn, x, residual, stats_ready=False):
     """BatchNorm (no inline relu) -> residual add+relu, the block-tail
-    pattern (resnet.hpp:41-48), with the BN's BACKWARD stats fused into
-    the add_relu backward kernel when eligible: the standalone
-    bwd_stats pass over (da, x) disappears. `bn` is the BatchNorm2d
-    module; eligibility = native path + training + flat-space grad views
-    + C % 8 == 0, C <= 1024 (kernel constraints)."""
+    pattern (resnet.hpp:41-48). When eligible (native path + training +
+    flat-space grad views + C % 8 == 0, C <= 1024) the whole tail runs as
+    ONE fused autograd node (_BnTail): forward normalize+add+relu in one
+    kernel, backward gating + BN stats in one kernel, then BN dx. `bn`
+    is the BatchNorm2d module (running stats updated; _nbt mirrored)."""
     fuse = (use_native(x) and bn.training
             and bn.weight.grad is not None and bn.bias.grad is not None
             and bn.num_features % 8 == 0 and bn.num_features <= 1024)
-    out = bn(x, stats_ready=stats_ready, stats_external=fuse)
     if fuse:
-        xb, mean, invstd = bn._bwd_stash
-        return _AddReLU.apply(out, residual, xb, mean, invstd,
-                              bn.weight.grad, bn.bias.grad)
+        bn._nbt += 1
+        have_stats = (stats_ready and bn.training
+                      and bn.num_features % 64 == 0)
+        return _BnTail.apply(x, residual, bn.weight, bn.bias,
+                             bn.running_mean, bn.running_var, bn.momentum,
+                             bn.eps, have_stats, bn.weight.grad,
+                             bn.bias.grad, bn)
+    out = bn(x, stats_ready=stats_ready)
     return add_relu(out, residual)
 
 
