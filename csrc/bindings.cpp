@@ -39,7 +39,12 @@ torch::Tensor gemm_bias(torch::Tensor, torch::Tensor, torch::Tensor, bool);
 torch::Tensor conv2d_fwd(torch::Tensor, torch::Tensor, torch::Tensor, long,
                          long, bool);
 torch::Tensor conv2d_dgrad(torch::Tensor, torch::Tensor, long, long, long,
-                           long);
+                           long, c10::optional<torch::Tensor>,
+                           c10::optional<torch::Tensor>,
+                           c10::optional<torch::Tensor>,
+                           c10::optional<torch::Tensor>,
+                           c10::optional<torch::Tensor>,
+                           c10::optional<torch::Tensor>);
 torch::Tensor conv2d_wgrad(torch::Tensor, torch::Tensor, long, long, long,
                            long);
 void conv2d_wgrad_into(torch::Tensor, torch::Tensor, torch::Tensor, long,
@@ -102,7 +107,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv2d_fwd", &eg::conv2d_fwd, py::arg("x"), py::arg("w"),
         py::arg("bias"), py::arg("stride"), py::arg("pad"),
         py::arg("collect_bn_stats") = false);
-  m.def("conv2d_dgrad", &eg::conv2d_dgrad);
+  m.def("conv2d_dgrad", &eg::conv2d_dgrad, py::arg("dy"), py::arg("wt"),
+        py::arg("stride"), py::arg("pad"), py::arg("H"), py::arg("W"),
+        py::arg("bs_y1") = c10::nullopt, py::arg("bs_x1") = c10::nullopt,
+        py::arg("bs_mean") = c10::nullopt,
+        py::arg("bs_invstd") = c10::nullopt,
+        py::arg("bs_dgamma") = c10::nullopt,
+        py::arg("bs_dbeta") = c10::nullopt);
   m.def("conv2d_wgrad", &eg::conv2d_wgrad);
   m.def("conv2d_wgrad_into", &eg::conv2d_wgrad_into);
   m.def("bn_fwd", &eg::bn_fwd, py::arg("x"), py::arg("gamma"), py::arg("beta"),

@@ -140,12 +140,27 @@ __device__ __forceinline__ void row8(const bf16* g, long rows, long cols,
 // the NEXT tile's global loads are issued right after this tile's LDS
 // write, so HBM latency hides under the MFMA phase.
 
-template <int MODE, bool FAST, int TBM, bool STATS = false>
+// backward-stats fusion pack (MODE 1 + BST): while writing dx — which IS
+// the downstream-in-backward BN's incoming dy — accumulate that BN's
+// sum(dy') and sum(dy' * xhat) into its grad views. y1 = the BN's output
+// (this conv's forward input, used for the relu gating), x1 = the BN's
+// input, mean/invstd its saved batch stats. Mirrors the MODE-0 STATS
+// forward fusion; the BN's backward then skips its standalone stats pass.
+struct BPack {
+  const bf16* y1;
+  const bf16* x1;
+  const float* mean;
+  const float* invstd;
+  float* sum_dy;
+  float* sum_dyx;
+};
+
+template <int MODE, bool FAST, int TBM, bool STATS = false, bool BST = false>
 __global__ __launch_bounds__(TBM * 4) void conv_mm_kernel(
     const bf16* __restrict__ Asrc, const bf16* __restrict__ B,
     const float* __restrict__ bias, bf16* __restrict__ out, Geom g,
     long M, long N, long RED, int has_bias,
-    float* __restrict__ stats_ws = nullptr) {
+    float* __restrict__ stats_ws = nullptr, BPack bp = BPack{}) {
   constexpr int THREADS = TBM * 4;
   constexpr int BROWS = 512 / THREADS;  // B-staging rows per thread
   // ONE shared allocation (guide §5.4 trap 4a: a second __shared__ object
@@ -156,11 +171,12 @@ __global__ __launch_bounds__(TBM * 4) void conv_mm_kernel(
   // barrier per k-tile instead of two. TBM=128 keeps a single buffer —
   // doubling its 27.6 KB LDS halves occupancy and measured net-slower.
   constexpr bool DB = (TBM == 64);
+  constexpr bool ACC = STATS || BST;   // either stats epilogue form
   constexpr int BUFSZ = TBM * CLDK + CBN * CLDK;
-  __shared__ __bf16 smem[(DB ? 2 : 1) * BUFSZ + (STATS ? 2 * CBN * 2 : 0)];
+  __shared__ __bf16 smem[(DB ? 2 : 1) * BUFSZ + (ACC ? 2 * CBN * 2 : 0)];
   float* s_sum = reinterpret_cast<float*>(smem + (DB ? 2 : 1) * BUFSZ);
   float* s_sq = s_sum + CBN;
-  if (STATS && threadIdx.x < CBN) {
+  if (ACC && threadIdx.x < CBN) {
     s_sum[threadIdx.x] = 0.f;
     s_sq[threadIdx.x] = 0.f;
   }
@@ -372,6 +388,11 @@ __global__ __launch_bounds__(TBM * 4) void conv_mm_kernel(
     long nn = n0 + wc * 32 + fj * 16 + cn;
     if (nn >= N) continue;
     float bv = has_bias ? bias[nn] : 0.f;
+    float bm = 0.f, bis = 0.f;
+    if (BST) {
+      bm = bp.mean[nn];
+      bis = bp.invstd[nn];
+    }
     float psum = 0.f, psq = 0.f;
 #pragma unroll
     for (int fi = 0; fi < 2; ++fi) {
@@ -385,9 +406,18 @@ __global__ __launch_bounds__(TBM * 4) void conv_mm_kernel(
           psum += v;
           psq += v * v;
         }
+        if (BST) {
+          // bf16 round-trip so the sums equal the unfused path, which
+          // reads the stored bf16 dx
+          const long o = mm * N + nn;
+          float gg = b2f(bp.y1[o]) > 0.f ? b2f(f2b(v)) : 0.f;
+          float xh = (b2f(bp.x1[o]) - bm) * bis;
+          psum += gg;
+          psq += gg * xh;
+        }
       }
     }
-    if (STATS) {
+    if (ACC) {
       // lanes l, l+16, l+32, l+48 hold the same output channel: reduce
       // across the wave first so only 16 lanes touch the LDS accumulators
       psum += __shfl_down(psum, 32);
@@ -408,6 +438,15 @@ __global__ __launch_bounds__(TBM * 4) void conv_mm_kernel(
         atomicAdd(&stats_ws[n0 + threadIdx.x], s_sum[threadIdx.x]);
         atomicAdd(&stats_ws[N + n0 + threadIdx.x], s_sq[threadIdx.x]);
       }
+    }
+  }
+  if (BST) {
+    __syncthreads();
+    if (threadIdx.x < CBN && n0 + threadIdx.x < N) {
+      if (s_sum[threadIdx.x] != 0.f)
+        atomicAdd(&bp.sum_dy[n0 + threadIdx.x], s_sum[threadIdx.x]);
+      if (s_sq[threadIdx.x] != 0.f)
+        atomicAdd(&bp.sum_dyx[n0 + threadIdx.x], s_sq[threadIdx.x]);
     }
   }
 }
@@ -1112,7 +1151,7 @@ torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor bias,
     fn<<<grid, (unsigned)(tbm * 4), 0, cur_stream()>>>(
         (const bf16*)x.data_ptr(), (const bf16*)w.data_ptr(),
         has_bias ? bias.data_ptr<float>() : nullptr, (bf16*)y.data_ptr(), g,
-        M, g.K, RED, has_bias ? 1 : 0, ws);
+        M, g.K, RED, has_bias ? 1 : 0, ws, conv::BPack{});
   } else {
     auto* fn = fast ? (narrow ? conv::conv_mm_kernel<0, true, 64>
                               : conv::conv_mm_kernel<0, true, 128>)
@@ -1121,14 +1160,23 @@ torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor bias,
     fn<<<grid, (unsigned)(tbm * 4), 0, cur_stream()>>>(
         (const bf16*)x.data_ptr(), (const bf16*)w.data_ptr(),
         has_bias ? bias.data_ptr<float>() : nullptr, (bf16*)y.data_ptr(), g,
-        M, g.K, RED, has_bias ? 1 : 0, nullptr);
+        M, g.K, RED, has_bias ? 1 : 0, nullptr, conv::BPack{});
   }
   return y;
 }
 
 // dy [N,Ho,Wo,K] bf16; wt [C,R,S,K] bf16 (w.permute(3,1,2,0)) -> dx [N,H,W,C]
+// Optional bs_* pack: fuse the upstream BN's backward stats into the dx
+// epilogue (see BPack). Only taken on the FAST stride-1 path; the caller
+// gates eligibility.
 torch::Tensor conv2d_dgrad(torch::Tensor dy, torch::Tensor wt, long stride,
-                           long pad, long H, long W) {
+                           long pad, long H, long W,
+                           c10::optional<torch::Tensor> bs_y1,
+                           c10::optional<torch::Tensor> bs_x1,
+                           c10::optional<torch::Tensor> bs_mean,
+                           c10::optional<torch::Tensor> bs_invstd,
+                           c10::optional<torch::Tensor> bs_dgamma,
+                           c10::optional<torch::Tensor> bs_dbeta) {
   CHECK_IN(dy); CHECK_IN(wt);
   int C = (int)wt.size(0), R = (int)wt.size(1), S = (int)wt.size(2);
   int K = (int)wt.size(3);
@@ -1192,13 +1240,27 @@ torch::Tensor conv2d_dgrad(torch::Tensor dy, torch::Tensor wt, long stride,
   bool narrow = blocks128 < 384;
   long tbm = narrow ? 64 : 128;
   dim3 grid(ceil_div(M, tbm), ceil_div(C, conv::CBN));
+  const bool bst = fast && bs_y1.has_value();
+  if (bst) {
+    TORCH_CHECK(C % 64 == 0, "bwd-stats fusion requires C % 64 == 0");
+    conv::BPack bp{(const bf16*)bs_y1->data_ptr(),
+                   (const bf16*)bs_x1->data_ptr(),
+                   bs_mean->data_ptr<float>(), bs_invstd->data_ptr<float>(),
+                   bs_dbeta->data_ptr<float>(), bs_dgamma->data_ptr<float>()};
+    auto* fn = narrow ? conv::conv_mm_kernel<1, true, 64, false, true>
+                      : conv::conv_mm_kernel<1, true, 128, false, true>;
+    fn<<<grid, (unsigned)(tbm * 4), 0, cur_stream()>>>(
+        (const bf16*)dy.data_ptr(), (const bf16*)wt.data_ptr(), nullptr,
+        (bf16*)dx.data_ptr(), g, M, C, RED, 0, nullptr, bp);
+    return dx;
+  }
   auto* fn = fast ? (narrow ? conv::conv_mm_kernel<1, true, 64>
                             : conv::conv_mm_kernel<1, true, 128>)
                   : (narrow ? conv::conv_mm_kernel<1, false, 64>
                             : conv::conv_mm_kernel<1, false, 128>);
   fn<<<grid, (unsigned)(tbm * 4), 0, cur_stream()>>>(
       (const bf16*)dy.data_ptr(), (const bf16*)wt.data_ptr(), nullptr,
-      (bf16*)dx.data_ptr(), g, M, C, RED, 0, nullptr);
+      (bf16*)dx.data_ptr(), g, M, C, RED, 0, nullptr, conv::BPack{});
   return dx;
 }
 

@@ -38,14 +38,15 @@ class Conv2d(nn.Module):
             bound = 1.0 / math.sqrt(fan_in)
             nn.init.uniform_(self.bias, -bound, bound)
 
-    def forward(self, x, bn_stats: bool = False):
+    def forward(self, x, bn_stats: bool = False, dgrad_stats_bn=None):
         # fused-BN-stats epilogue requires out_ch % 64 == 0 (csrc/conv.hip);
         # the paired BatchNorm downgrades stats_ready by the same condition
         return O.conv2d(x, self.weight, self.bias, self.stride, self.padding,
                         bn_stats=bn_stats and O.use_native(x) and self.training
                         and self.out_ch % 64 == 0,
                         wk=getattr(self, "_shadow_wk", None),
-                        wt=getattr(self, "_shadow_wt", None))
+                        wt=getattr(self, "_shadow_wt", None),
+                        dgrad_stats_bn=dgrad_stats_bn)
 
     def extra_repr(self):
         return (f"{self.in_ch}, {self.out_ch}, k={self.kernel_size}, "
@@ -80,14 +81,16 @@ class BatchNorm2d(nn.Module):
     def sync_buffers_after_load(self):
         self._nbt = int(self.num_batches_tracked.item())
 
-    def forward(self, x, fuse_relu: bool = False, stats_ready: bool = False):
+    def forward(self, x, fuse_relu: bool = False, stats_ready: bool = False,
+                stats_consumer: bool = False):
         if self.training:
             self._nbt += 1
         return O.batch_norm(x, self.weight, self.bias, self.running_mean,
                             self.running_var, self.training, self.momentum,
                             self.eps, fuse_relu,
                             stats_ready and O.use_native(x) and self.training
-                            and self.num_features % 64 == 0)
+                            and self.num_features % 64 == 0,
+                            stash_module=self if stats_consumer else None)
 
 
 class Linear(nn.Module):

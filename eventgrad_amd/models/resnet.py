@@ -54,13 +54,17 @@ class BasicBlock(nn.Module):
                             if downsample else None)
 
     def forward(self, x):
+        # bn1 -> conv2 is single-consumer: conv2's dgrad epilogue can
+        # accumulate bn1's backward stats (ops/functional.py)
+        f1 = O.can_fuse_dgrad_stats(self.bn1, self.conv2, x)
         out = self.bn1(self.conv1(x, bn_stats=True), fuse_relu=True,
-                       stats_ready=True)
+                       stats_ready=True, stats_consumer=f1)
         # residual FIRST: the downsampler's conv+BN use the same shared
         # per-channel-count stats workspace as the conv2->bn2 pair below —
         # interleaving them corrupts both stats streams
         residual = self.downsampler(x) if self.downsampler is not None else x
-        out = self.conv2(out, bn_stats=True)
+        out = self.conv2(out, bn_stats=True,
+                         dgrad_stats_bn=self.bn1 if f1 else None)
         # block tail: bn2 -> add+relu with bn2's backward stats fused into
         # the add_relu backward kernel (ops/functional.py::bn_add_relu)
         return O.bn_add_relu(self.bn2, out, residual, stats_ready=True)
@@ -81,12 +85,16 @@ class BottleNeck(nn.Module):
                             if downsample else None)
 
     def forward(self, x):
+        f1 = O.can_fuse_dgrad_stats(self.bn1, self.conv2, x)
         out = self.bn1(self.conv1(x, bn_stats=True), fuse_relu=True,
-                       stats_ready=True)
-        out = self.bn2(self.conv2(out, bn_stats=True), fuse_relu=True,
-                       stats_ready=True)
+                       stats_ready=True, stats_consumer=f1)
+        f2 = O.can_fuse_dgrad_stats(self.bn2, self.conv3, out)
+        out = self.bn2(self.conv2(out, bn_stats=True,
+                                  dgrad_stats_bn=self.bn1 if f1 else None),
+                       fuse_relu=True, stats_ready=True, stats_consumer=f2)
         residual = self.downsampler(x) if self.downsampler is not None else x
-        out = self.conv3(out, bn_stats=True)
+        out = self.conv3(out, bn_stats=True,
+                         dgrad_stats_bn=self.bn2 if f2 else None)
         return O.bn_add_relu(self.bn3, out, residual, stats_ready=True)
 
 

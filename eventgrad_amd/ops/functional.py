@@ -104,7 +104,8 @@ class _ConvNHWC(torch.autograd.Function):
     """
 
     @staticmethod
-    def forward(ctx, x, w, bias, stride, padding, bn_stats, wk, wt):
+    def forward(ctx, x, w, bias, stride, padding, bn_stats, wk, wt,
+                dgrad_stats_bn):
         core = native()
         if wk is None:
             wk = _w_krsc(w)
@@ -115,6 +116,16 @@ class _ConvNHWC(torch.autograd.Function):
         ctx.wgrad = w.grad if w.requires_grad else None
         ctx.bgrad = (bias.grad if (bias is not None and bias.requires_grad)
                      else None)
+        # dgrad-side BN-stats fusion: this conv's input x IS the upstream
+        # BN's relu output y1; the BN stashed (x1, mean, invstd) at its
+        # forward (stats_consumer=True) and its backward skips the
+        # standalone stats pass
+        if dgrad_stats_bn is not None:
+            x1, mean1, invstd1 = dgrad_stats_bn._bwd_stash
+            ctx.bs = (x1, mean1, invstd1, dgrad_stats_bn.weight.grad,
+                      dgrad_stats_bn.bias.grad)
+        else:
+            ctx.bs = None
         ctx.stride, ctx.padding = stride, padding
         ctx.has_bias = bias is not None
         ctx.hw = (x.shape[1], x.shape[2])
@@ -130,8 +141,16 @@ class _ConvNHWC(torch.autograd.Function):
         if ctx.needs_input_grad[0]:
             # dgrad weight layout: [C,R,S,K] from [K,R,S,C]
             w_crsk = ctx.wt if ctx.wt is not None else core.krsc_to_crsk(wk)
-            dx = core.conv2d_dgrad(dy, w_crsk, ctx.stride, ctx.padding,
-                                   ctx.hw[0], ctx.hw[1])
+            if ctx.bs is not None:
+                x1, mean1, invstd1, ggrad, bgrad = ctx.bs
+                dx = core.conv2d_dgrad(dy, w_crsk, ctx.stride, ctx.padding,
+                                       ctx.hw[0], ctx.hw[1], bs_y1=x,
+                                       bs_x1=x1, bs_mean=mean1,
+                                       bs_invstd=invstd1, bs_dgamma=ggrad,
+                                       bs_dbeta=bgrad)
+            else:
+                dx = core.conv2d_dgrad(dy, w_crsk, ctx.stride, ctx.padding,
+                                       ctx.hw[0], ctx.hw[1])
         if ctx.needs_input_grad[1]:
             if ctx.wgrad is not None:
                 core.conv2d_wgrad_into(x, dy, ctx.wgrad, R, S,
@@ -145,19 +164,21 @@ class _ConvNHWC(torch.autograd.Function):
                 core.channel_sum_into(dy, ctx.bgrad)
             else:
                 db = core.channel_sum(dy)
-        return dx, dw, db, None, None, None, None, None
+        return dx, dw, db, None, None, None, None, None, None
 
 
 def conv2d(x, w, bias=None, stride=1, padding=0, bn_stats=False,
-           wk=None, wt=None):
+           wk=None, wt=None, dgrad_stats_bn=None):
     """bn_stats: fuse the following training-mode BatchNorm's batch-stats
     accumulation into this conv's epilogue (pair with
     batch_norm(..., stats_ready=True)). wk/wt: persistent bf16 KRSC/CRSK
     shadow views maintained by FlatParamSpace.refresh_shadows (skips the
-    per-use layout transforms)."""
+    per-use layout transforms). dgrad_stats_bn: the upstream BatchNorm2d
+    whose BACKWARD stats this conv's dgrad epilogue accumulates (the BN
+    ran with stats_consumer=True)."""
     if use_native(x):
         return _ConvNHWC.apply(x, w, bias, int(stride), int(padding),
-                               bool(bn_stats), wk, wt)
+                               bool(bn_stats), wk, wt, dgrad_stats_bn)
     return F.conv2d(x, w, bias, stride=stride, padding=padding)
 
 
@@ -168,12 +189,17 @@ def conv2d(x, w, bias=None, stride=1, padding=0, bn_stats=False,
 class _BatchNormNHWC(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, gamma, beta, running_mean, running_var, training,
-                momentum, eps, fuse_relu, stats_ready):
+                momentum, eps, fuse_relu, stats_ready, stash_module):
         core = native()
         y, save_mean, save_invstd = core.bn_fwd(
             x, gamma.detach(), beta.detach(), running_mean, running_var,
             momentum, eps, training, fuse_relu, stats_ready)
         ctx.save_for_backward(x, gamma, save_mean, save_invstd, y)
+        # stats_consumer: the consuming conv's dgrad epilogue accumulates
+        # this BN's backward stats into the grad views (skip them here)
+        ctx.stats_external = stash_module is not None
+        if stash_module is not None:
+            stash_module._bwd_stash = (x, save_mean, save_invstd)
         # direct-grad views (see _ConvNHWC): the bwd reduction buffers ARE
         # dgamma/dbeta, so pointing them at the pre-zeroed flat-grad views
         # writes the gradients in place with zero extra kernels
@@ -194,22 +220,24 @@ class _BatchNormNHWC(torch.autograd.Function):
                                    save_invstd, gamma.detach(), y,
                                    ctx.fuse_relu, ctx.training,
                                    dgamma_out=ctx.ggrad,
-                                   dbeta_out=ctx.bgrad)
+                                   dbeta_out=ctx.bgrad,
+                                   stats_ready=ctx.stats_external)
             return (dx, None, None, None, None, None, None, None, None,
-                    None)
+                    None, None)
         dx, dgamma, dbeta = core.bn_bwd(dy.contiguous(), x, save_mean,
                                         save_invstd, gamma.detach(), y,
                                         ctx.fuse_relu, ctx.training)
         return (dx, dgamma, dbeta, None, None, None, None, None, None,
-                None)
+                None, None)
 
 
 def batch_norm(x, gamma, beta, running_mean, running_var, training,
-               momentum=0.1, eps=1e-5, fuse_relu=False, stats_ready=False):
+               momentum=0.1, eps=1e-5, fuse_relu=False, stats_ready=False,
+               stash_module=None):
     if use_native(x):
         return _BatchNormNHWC.apply(x, gamma, beta, running_mean, running_var,
                                     training, momentum, eps, fuse_relu,
-                                    stats_ready and training)
+                                    stats_ready and training, stash_module)
     y = F.batch_norm(x, running_mean, running_var, gamma, beta, training,
                      momentum, eps)
     return F.relu(y) if fuse_relu else y
@@ -289,6 +317,16 @@ class _BnTail(torch.autograd.Function):
                                dbeta_out=ctx.bgrad, stats_ready=True)
         return (dx, da, None, None, None, None, None, None, None, None,
                 None, None)
+
+
+def can_fuse_dgrad_stats(bn, conv, x) -> bool:
+    """Eligibility for fusing `bn`'s backward stats into `conv`'s dgrad
+    epilogue (bn -> conv single-consumer chain inside a block)."""
+    return (use_native(x) and bn.training
+            and bn.weight.grad is not None and bn.bias.grad is not None
+            and bn.num_features % 64 == 0
+            and conv.stride == 1 and conv.out_ch % 8 == 0
+            and conv.kernel_size * conv.kernel_size <= 32)
 
 
 def bn_add_relu(bn, x, residual, stats_ready=False):

@@ -385,3 +385,44 @@ def test_bn_tail_fusion_grads_exact():
                                dbeta_ref.cpu().numpy(), rtol=1e-4, atol=1e-2)
     np.testing.assert_allclose(block.bn2.weight.grad.cpu().numpy(),
                                dgamma_ref.cpu().numpy(), rtol=1e-3, atol=5e-2)
+
+
+def test_dgrad_bn_stats_fusion_exact():
+    """bn1 -> conv2 fusion: the conv dgrad epilogue's accumulated BN stats
+    must equal fp32 math on the retained upstream gradient."""
+    from eventgrad_amd.models.resnet import BasicBlock
+    from eventgrad_amd.ops import functional as O
+    from eventgrad_amd.parallel.flat import FlatParamSpace
+
+    dev = torch.device("cuda")
+    torch.manual_seed(11)
+    block = BasicBlock(64, 64).to(dev)
+    block.train()
+    space = FlatParamSpace(block, dev)
+
+    x = torch.randn(4, 8, 8, 64, device=dev) \
+        .permute(0, 2, 3, 1).contiguous().to(torch.bfloat16)
+    # replicate the block body with a retained intermediate
+    space.zero_grad()
+    assert O.can_fuse_dgrad_stats(block.bn1, block.conv2, x)
+    out1 = block.bn1(block.conv1(x, bn_stats=True), fuse_relu=True,
+                     stats_ready=True, stats_consumer=True)
+    out1.retain_grad()
+    out2 = block.conv2(out1, bn_stats=True, dgrad_stats_bn=block.bn1)
+    out = O.bn_add_relu(block.bn2, out2, x, stats_ready=True)
+    out.float().sum().backward()
+    torch.cuda.synchronize()
+
+    dy1 = out1.grad.float()                      # conv2's dgrad output
+    gated = torch.where(out1.detach().float() > 0, dy1,
+                        torch.zeros(1, device=dev))
+    x1, mean, invstd = block.bn1._bwd_stash
+    xh = (x1.detach().float() - mean) * invstd
+    C = 64
+    dbeta_ref = gated.reshape(-1, C).sum(0)
+    dgamma_ref = (gated * xh).reshape(-1, C).sum(0)
+    np.testing.assert_allclose(block.bn1.bias.grad.cpu().numpy(),
+                               dbeta_ref.cpu().numpy(), rtol=1e-3, atol=1e-2)
+    np.testing.assert_allclose(block.bn1.weight.grad.cpu().numpy(),
+                               dgamma_ref.cpu().numpy(), rtol=1e-3,
+                               atol=5e-2)
