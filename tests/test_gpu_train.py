@@ -400,8 +400,8 @@ def test_dgrad_bn_stats_fusion_exact():
     block.train()
     space = FlatParamSpace(block, dev)
 
-    x = torch.randn(4, 8, 8, 64, device=dev) \
-        .permute(0, 2, 3, 1).contiguous().to(torch.bfloat16)
+    x = torch.randn(4, 64, 8, 8, device=dev) \
+        .permute(0, 2, 3, 1).contiguous().to(torch.bfloat16)  # NHWC
     # replicate the block body with a retained intermediate
     space.zero_grad()
     assert O.can_fuse_dgrad_stats(block.bn1, block.conv2, x)
