@@ -70,13 +70,17 @@ def main():
     ap.add_argument("--train-samples", type=int, default=2048)
     ap.add_argument("--epochs", type=int, default=8)
     ap.add_argument("--force", action="store_true")
+    ap.add_argument("--arms", nargs="+", default=None,
+                    help="subset of arm names to run (default: all)")
     args = ap.parse_args()
     if args.noise is None:
         # calibrated so decent lands ~85-95% test acc (non-saturating)
         args.noise = 1.35 if args.experiment == "mnist" else 1.2
 
     rows = {}
-    for name, extra in ARMS:
+    arms = [(n, e) for n, e in ARMS
+            if args.arms is None or n in args.arms]
+    for name, extra in arms:
         per_seed = []
         for seed in args.seeds:
             r = run_arm(args.experiment, seed, name, extra, args)
