@@ -10,7 +10,7 @@ compute_dtype value) — and records the loss/accuracy trajectories, so the
 bf16 headline throughput number carries a precision-parity citation.
 
 Run on a GPU box:  python benchmarks/precision_parity.py [--epochs 8]
-Writes benchmarks/out/precision_parity.json.
+Writes benchmarks/precision_parity.json.
 """
 
 from __future__ import annotations
@@ -67,8 +67,7 @@ def main():
     out["final_acc_curves_max_gap"] = round(max(
         abs(a - b) for a, b in zip(bf16["epoch_train_acc"],
                                    fp32["epoch_train_acc"])), 3)
-    os.makedirs(os.path.join(HERE, "out"), exist_ok=True)
-    path = os.path.join(HERE, "out", "precision_parity.json")
+    path = os.path.join(HERE, "precision_parity.json")
     with open(path, "w") as f:
         json.dump(out, f, indent=2)
     print(json.dumps(out, indent=2))
