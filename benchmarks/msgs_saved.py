@@ -47,6 +47,9 @@ def build_cfg(experiment: str, mode: str, args):
     cfg.data.synthetic_noise = args.noise
     cfg.data.label_noise = args.label_noise
     cfg.seed = args.seed
+    if args.global_batch:
+        cfg.data.global_batch = args.global_batch
+        cfg.data.batch_size = args.global_batch
     cfg.mode = mode
     cfg.trigger.adaptive = not args.constant
     cfg.trigger.horizon = args.horizon
@@ -95,6 +98,8 @@ def main():
                     help="fraction of train labels flipped (test clean); "
                          "holds accuracy in a non-saturating band")
     ap.add_argument("--seed", type=int, default=0)
+    ap.add_argument("--global-batch", type=int, default=None,
+                    help="override the global batch (more passes per epoch)")
     ap.add_argument("--horizon", type=float, default=1.01)
     ap.add_argument("--constant", action="store_true",
                     help="use static threshold instead of adaptive")

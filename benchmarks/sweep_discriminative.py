@@ -42,7 +42,8 @@ ARMS = [
 
 
 def run_arm(experiment, seed, name, extra, args):
-    outdir = os.path.join(HERE, "out_disc", f"{experiment}_s{seed}_{name}")
+    outdir = os.path.join(HERE, "out_disc",
+                          f"{experiment}{args.tag}_s{seed}_{name}")
     mode = extra[1]
     result = os.path.join(outdir, f"{mode}.json")
     if os.path.exists(result) and not args.force:
@@ -54,6 +55,8 @@ def run_arm(experiment, seed, name, extra, args):
            "--label-noise", str(args.label_noise),
            "--train-samples", str(args.train_samples),
            "--epochs", str(args.epochs), "--outdir", outdir] + extra
+    if args.global_batch:
+        cmd += ["--global-batch", str(args.global_batch)]
     subprocess.run(cmd, check=True, capture_output=True)
     with open(result) as f:
         return json.load(f)
@@ -72,6 +75,10 @@ def main():
     ap.add_argument("--force", action="store_true")
     ap.add_argument("--arms", nargs="+", default=None,
                     help="subset of arm names to run (default: all)")
+    ap.add_argument("--global-batch", type=int, default=None)
+    ap.add_argument("--tag", default="",
+                    help="suffix for result dirs / sweep json (distinct "
+                         "configs do not share caches)")
     args = ap.parse_args()
     if args.noise is None:
         # calibrated so decent lands ~85-95% test acc (non-saturating)
@@ -109,12 +116,14 @@ def main():
         "world": args.world,
         "config": {"noise": args.noise, "label_noise": args.label_noise,
                    "train_samples": args.train_samples,
-                   "epochs": args.epochs, "seeds": args.seeds},
+                   "epochs": args.epochs, "seeds": args.seeds,
+                   "global_batch": args.global_batch},
         "decent_seed_spread": round(spread, 2),
         "arms": rows,
     }
     os.makedirs(os.path.join(HERE, "out_disc"), exist_ok=True)
-    path = os.path.join(HERE, "out_disc", f"{args.experiment}_sweep.json")
+    path = os.path.join(HERE, "out_disc",
+                        f"{args.experiment}{args.tag}_sweep.json")
     with open(path, "w") as f:
         json.dump(out, f, indent=2)
     print(json.dumps(out, indent=2))
