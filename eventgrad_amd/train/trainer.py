@@ -79,9 +79,16 @@ class Trainer:
         graph = None
         use_graph = cfg.hip_graph and can_graph(self.model, self.device)
         t0 = time.perf_counter()
+        # device-resident accuracy accumulator: a per-pass .item() would
+        # sync the stream every pass, serializing host data prep against
+        # GPU compute (and letting the clocks idle between bursts) — the
+        # round-1 trainer-vs-bench throughput gap was mostly this
+        dev_correct = (torch.zeros(1, dtype=torch.int64, device=self.device)
+                       if self.device.type == "cuda" else None)
         for epoch in range(self.start_epoch, cfg.epochs + 1):
             correct = seen = 0
             last_loss = 0.0
+            loss = None
             for x, y in self.loader.epoch(epoch):
                 x = x.to(self.device, non_blocking=True)
                 y = y.to(self.device, non_blocking=True)
@@ -102,8 +109,16 @@ class Trainer:
                     loss.backward()
                 self.engine.after_backward()
                 self.engine.step()
-                correct += O.accuracy_count(logits.detach(), y)
+                if dev_correct is not None:
+                    dev_correct += (logits.detach().float().argmax(dim=1)
+                                    == y).sum()
+                else:
+                    correct += O.accuracy_count(logits.detach(), y)
                 seen += x.shape[0]
+            if dev_correct is not None:
+                correct = int(dev_correct.item())  # one sync per epoch
+                dev_correct.zero_()
+            if loss is not None:
                 last_loss = float(loss.detach())
             acc = 100.0 * correct / max(seen, 1)
             self.metrics.epoch_train_acc.append(round(acc, 4))
