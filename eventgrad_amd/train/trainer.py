@@ -85,13 +85,35 @@ class Trainer:
         # round-1 trainer-vs-bench throughput gap was mostly this
         dev_correct = (torch.zeros(1, dtype=torch.int64, device=self.device)
                        if self.device.type == "cuda" else None)
+        # double-buffered pinned H2D staging: a pageable .to() measured
+        # ~5 ms/batch; pinned + non_blocking overlaps the DMA with compute.
+        # Two buffers + events so the host never overwrites a buffer whose
+        # DMA is still in flight.
+        pin = [None, None]
+        pin_ev = [None, None]
+        pi = 0
         for epoch in range(self.start_epoch, cfg.epochs + 1):
             correct = seen = 0
             last_loss = 0.0
             loss = None
             for x, y in self.loader.epoch(epoch):
-                x = x.to(self.device, non_blocking=True)
-                y = y.to(self.device, non_blocking=True)
+                if dev_correct is not None:
+                    if pin[pi] is None or pin[pi][0].shape != x.shape:
+                        pin[pi] = (torch.empty_like(x).pin_memory(),
+                                   torch.empty_like(y).pin_memory())
+                        pin_ev[pi] = torch.cuda.Event()
+                        pin_ev[pi].record()
+                    pin_ev[pi].synchronize()  # prior DMA from buffer done
+                    px, py = pin[pi]
+                    px.copy_(x)
+                    py.copy_(y)
+                    x = px.to(self.device, non_blocking=True)
+                    y = py.to(self.device, non_blocking=True)
+                    pin_ev[pi].record()
+                    pi ^= 1
+                else:
+                    x = x.to(self.device, non_blocking=True)
+                    y = y.to(self.device, non_blocking=True)
                 self.pass_num += 1
                 if use_graph and graph is None:
                     # capture before the first comm pass posts RCCL work
