@@ -105,8 +105,11 @@ class Trainer:
                         pin_ev[pi].record()
                     pin_ev[pi].synchronize()  # prior DMA from buffer done
                     px, py = pin[pi]
-                    px.copy_(x)
-                    py.copy_(y)
+                    # numpy memcpy: a torch CPU copy_ is thread-dispatch
+                    # bound on many-core hosts (same pathology as the
+                    # dataset gather — measured ~4x slower)
+                    px.numpy()[:] = x.numpy()
+                    py.numpy()[:] = y.numpy()
                     x = px.to(self.device, non_blocking=True)
                     y = py.to(self.device, non_blocking=True)
                     pin_ev[pi].record()
