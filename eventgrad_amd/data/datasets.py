@@ -54,6 +54,7 @@ class SyntheticImages:
         # trainer). Blocks stay cached: a full 50k CIFAR-shaped set is
         # ~600 MB host RAM once every block has been touched.
         self._cache: dict = {}
+        self._all: np.ndarray | None = None  # consolidated (see batch)
 
     def __len__(self):
         return self.n
@@ -75,18 +76,30 @@ class SyntheticImages:
 
     def __getitem__(self, i: int):
         i = int(i)
-        x = torch.from_numpy(
-            self._block(i // self.BLOCK)[i % self.BLOCK].copy())
+        if self._all is not None:
+            x = torch.from_numpy(self._all[i].copy())
+        else:
+            x = torch.from_numpy(
+                self._block(i // self.BLOCK)[i % self.BLOCK].copy())
         return x, int(self.labels[i])
 
     def batch(self, idx: list) -> Tuple[torch.Tensor, torch.Tensor]:
         idx = np.asarray(idx, dtype=np.int64)
+        ys = torch.from_numpy(self.labels[idx])
+        nblocks = (self.n + self.BLOCK - 1) // self.BLOCK
+        if self._all is None and len(self._cache) == nblocks:
+            # every block generated: consolidate once so a shuffled batch
+            # is a single np.take instead of ~n/BLOCK per-block gathers
+            self._all = np.concatenate(
+                [self._cache[b] for b in range(nblocks)])
+            self._cache.clear()
+        if self._all is not None:
+            return torch.from_numpy(self._all[idx]), ys
         xs = np.empty((len(idx),) + self.shape, dtype=np.float32)
         blocks = idx // self.BLOCK
         for b in np.unique(blocks):
             sel = np.nonzero(blocks == b)[0]
             xs[sel] = self._block(int(b))[idx[sel] % self.BLOCK]
-        ys = torch.from_numpy(self.labels[idx])
         return torch.from_numpy(xs), ys
 
 
