@@ -183,3 +183,34 @@ def test_jpeg_dir_cifar10(tmp_path):
     cfg = DataConfig(dataset="cifar10", data_path=str(tmp_path))
     d2 = build_dataset(cfg, train=False)
     assert isinstance(d2, JpegDirCifar10) and len(d2) == 20
+
+
+def test_trainer_on_jpeg_dir_dataset(tmp_path):
+    """End-to-end: the reference's JPEG-dir CIFAR contract feeds the
+    Trainer (serial, 1 epoch) through build_dataset auto-detection."""
+    import numpy as np
+    from PIL import Image
+    from eventgrad_amd.config import DataConfig, OptimConfig, RunConfig
+    from eventgrad_amd.data.datasets import CIFAR10_CLASSES
+    from eventgrad_amd.train.trainer import Trainer
+
+    rng = np.random.default_rng(1)
+    for split, n in (("train", 8), ("test", 4)):
+        for j, cls in enumerate(CIFAR10_CLASSES):
+            d = tmp_path / split / cls
+            d.mkdir(parents=True)
+            for i in range(n):
+                base = np.zeros((32, 32, 3), np.uint8)
+                base[..., j % 3] = 150 + 10 * (j // 3)
+                base += rng.integers(0, 20, base.shape).astype(np.uint8)
+                Image.fromarray(base).save(d / f"{i:04d}.jpg", quality=92)
+
+    cfg = RunConfig(mode="serial", model="lenet5", epochs=1, device="cpu",
+                    data=DataConfig(dataset="cifar10",
+                                    data_path=str(tmp_path), batch_size=16),
+                    optim=OptimConfig(lr=1e-3), eval_at_end=True)
+    tr = Trainer(cfg)
+    m = tr.train()
+    assert m.total_passes == 5            # 80 train images / 16
+    assert np.isfinite(m.final_train_loss)
+    assert m.test_accuracy is not None    # eval over the 40 test JPEGs
