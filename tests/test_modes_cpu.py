@@ -120,3 +120,18 @@ def test_checkpoint_resume_bn_model(tmp_path):
         if d["ref_eval"] is not None:  # rank 0 evaluated
             assert d["res_eval"] == d["ref_eval"]
             assert abs(d["res_loss"] - d["ref_loss"]) < 1e-6
+
+
+def test_event_runs_world3_odd_ring(tmp_path):
+    """Odd ring: left != right for every rank and the ring is not
+    edge-symmetric — exercises mask/payload pairing beyond the
+    world-2 same-peer case and the even-ring world 4/8 tests."""
+    run_world(W.train_mode_worker, 3, "event", str(tmp_path),
+              '{"batch_size": 16, "n_train": 192}')
+    files = sorted(glob.glob(os.path.join(tmp_path, "event_r*.pt")))
+    assert len(files) == 3
+    outs = [torch.load(f, weights_only=False) for f in files]
+    for o in outs:
+        assert torch.isfinite(o["param"]).all()
+        assert torch.allclose(outs[0]["param"], o["param"], atol=1e-6)
+    assert 0.0 <= outs[0]["metrics"]["messages_saved_pct"] < 100.0
