@@ -10,8 +10,9 @@ The reference, under ``file_write == 1``, writes per pass one line per file
 We keep the same filenames and column meaning, but always write the 0/1
 new-msg flag (the reference omits the '0' case on MNIST recv — ambiguous to
 parse; dcifar10/spevent writes both, spevent.cpp:470-478, which is the format
-used here). A train{rank}.txt with per-epoch accuracy mirrors
-cent.cpp:80-92.
+used here). A train{rank}.txt carries per-epoch accuracy+loss, and
+values{rank}.txt carries the reference cent binary's exact per-epoch
+"<epoch>, <loss>" lines (cent.cpp:80-92,124).
 """
 
 from __future__ import annotations
@@ -25,6 +26,7 @@ class Tracer:
         self.fps = open(os.path.join(outdir, f"send{rank}.txt"), "w")
         self.fpr = open(os.path.join(outdir, f"recv{rank}.txt"), "w")
         self.fpt = open(os.path.join(outdir, f"train{rank}.txt"), "w")
+        self.fpv = open(os.path.join(outdir, f"values{rank}.txt"), "w")
 
     def send_line(self, norms, thres, fired):
         parts = [f"{n},  {t},  {int(f)},  "
@@ -39,7 +41,8 @@ class Tracer:
 
     def train_line(self, epoch: int, accuracy: float, loss: float):
         self.fpt.write(f"{epoch}, {accuracy}, {loss}\n")
+        self.fpv.write(f"{epoch}, {loss}\n")  # cent.cpp:124 format
 
     def close(self):
-        for f in (self.fps, self.fpr, self.fpt):
+        for f in (self.fps, self.fpr, self.fpt, self.fpv):
             f.close()

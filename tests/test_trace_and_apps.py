@@ -95,3 +95,15 @@ def test_cli_json_out(tmp_path):
     import json
     m = json.load(open(out))
     assert m["total_passes"] > 0
+
+
+def test_values_file_format(tmp_path):
+    """values{rank}.txt: the cent binary's per-epoch '<epoch>, <loss>'
+    lines (cent.cpp:80-92,124)."""
+    from eventgrad_amd.train.trace import Tracer
+    tr = Tracer(0, str(tmp_path))
+    tr.train_line(1, 50.0, 2.25)
+    tr.train_line(2, 75.0, 1.5)
+    tr.close()
+    lines = open(tmp_path / "values0.txt").read().splitlines()
+    assert lines == ["1, 2.25", "2, 1.5"]
