@@ -172,6 +172,12 @@ class CommEngine:
         # bf16 weight shadows the next forward/backward will read
         self.space.refresh_shadows()
 
+    def drain(self) -> None:
+        """Complete any in-flight lookahead comm (no-op for most engines).
+
+        Must run before the process group is destroyed even when the
+        closing consensus (finalize) is skipped (final_consensus=False)."""
+
     def finalize(self) -> dict:
         """Closing consensus: params <- allreduce/world (event.cpp:517-525).
 
@@ -349,12 +355,15 @@ class GossipEngine(CommEngine):
             self._trace_recv()
         self.K.avg3(self.space.param, self.inbox_left, self.inbox_right)
 
-    def finalize(self) -> dict:
+    def drain(self) -> None:
         # drain a lookahead mask exchange whose pass never ran (training
         # ended): peers posted symmetrically, so waiting completes it
         if self.transport is not None:
             self.transport.cancel_pending_masks()
         self._pending = None
+
+    def finalize(self) -> dict:
+        self.drain()
         return super().finalize()
 
     def _trace_recv(self):

@@ -168,6 +168,9 @@ class Trainer:
             stats = self.engine.finalize()
             self.metrics.num_events_total = stats["num_events_total"]
         else:
+            # no consensus, but in-flight lookahead comm must still drain
+            # before the process group can be torn down
+            self.engine.drain()
             self.metrics.num_events_total = self.engine.num_events * self.world
         if self.rank == 0:
             print(f"Training time - {self.metrics.train_time_s}", flush=True)
