@@ -67,7 +67,12 @@ def preflight(rank: int, world: int, device) -> int:
             engine.step()
             fired = getattr(engine, "_fired_l", None)
             fired_hist.append(-1 if fired is None else len(fired))
+        # modes share one process group and fixed tags: the trailing
+        # lookahead mask exchange must drain before the next mode posts
+        engine.drain()
         torch.cuda.synchronize()
+        if world > 1:
+            dist.barrier()
         digest = {
             "rank": rank,
             "param_sqnorm": round(float(space.param.square().sum()), 4),
@@ -190,6 +195,7 @@ def main() -> int:
         loss = step()
     barrier_sync()
     elapsed = time.perf_counter() - t0
+    engine.drain()  # complete the last step's lookahead mask exchange
 
     if world > 1:
         t = torch.tensor([elapsed], device=device)
